@@ -1,0 +1,168 @@
+"""GPT-2 example model (nanoGPT-style), written for the MI355X op stack.
+
+Capability parity with ``/root/reference/example/model.py:15-157`` (same
+architecture family and config fields: 12L/12H/768d GPT-2 small by default,
+vocab 50304, block 1024, bias=False, selectable attention backend, returns
+(logits, loss)). Differences by design:
+
+- attention, GELU and the loss go through tiny_deepspeed_amd.ops — on GPU
+  these are the hand-written CDNA4 kernels (fused causal attention,
+  elementwise GELU, fused cross-entropy); on CPU the torch fallbacks.
+- nn.Linear / nn.LayerNorm / nn.Embedding are used raw so the parallel
+  wrappers can swap them for strategy modules (SURVEY.md component #14).
+- presets for small/medium/large/xl support the BASELINE.json configs.
+"""
+
+import math
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+
+
+@dataclass
+class GPTConfig:
+    block_size: int = 1024
+    vocab_size: int = 50304
+    n_layer: int = 12
+    n_head: int = 12
+    n_embd: int = 768
+    dropout: float = 0.0
+    bias: bool = False
+    # "fused": CDNA4 flash-style kernel (composite torch on CPU);
+    # "math": composite torch attention everywhere.
+    attention: str = "fused"
+
+    @classmethod
+    def gpt2_small(cls, **kw):
+        return cls(n_layer=12, n_head=12, n_embd=768, **kw)
+
+    @classmethod
+    def gpt2_medium(cls, **kw):
+        return cls(n_layer=24, n_head=16, n_embd=1024, **kw)
+
+    @classmethod
+    def gpt2_large(cls, **kw):
+        return cls(n_layer=36, n_head=20, n_embd=1280, **kw)
+
+    @classmethod
+    def gpt2_xl(cls, **kw):
+        return cls(n_layer=48, n_head=25, n_embd=1600, **kw)
+
+    @classmethod
+    def named(cls, name, **kw):
+        presets = {
+            "gpt2-small": cls.gpt2_small,
+            "gpt2-medium": cls.gpt2_medium,
+            "gpt2-large": cls.gpt2_large,
+            "gpt2-xl": cls.gpt2_xl,
+        }
+        return presets[name](**kw)
+
+
+class CausalSelfAttention(nn.Module):
+    def __init__(self, config):
+        super().__init__()
+        assert config.n_embd % config.n_head == 0
+        self.n_head = config.n_head
+        self.head_dim = config.n_embd // config.n_head
+        self.attention = config.attention
+        self.dropout = config.dropout
+        self.c_attn = nn.Linear(config.n_embd, 3 * config.n_embd, bias=config.bias)
+        self.c_proj = nn.Linear(config.n_embd, config.n_embd, bias=config.bias)
+
+    def forward(self, x):
+        B, T, E = x.shape
+        qkv = self.c_attn(x)
+        q, k, v = qkv.split(E, dim=2)
+        # (B, T, E) -> (B, H, T, D)
+        q = q.view(B, T, self.n_head, self.head_dim).transpose(1, 2).contiguous()
+        k = k.view(B, T, self.n_head, self.head_dim).transpose(1, 2).contiguous()
+        v = v.view(B, T, self.n_head, self.head_dim).transpose(1, 2).contiguous()
+        if self.attention == "fused":
+            y = ops.causal_attention(q, k, v, dropout_p=self.dropout,
+                                     training=self.training)
+        elif self.attention == "math":
+            scale = 1.0 / math.sqrt(self.head_dim)
+            mask = torch.ones(T, T, dtype=torch.bool, device=x.device).tril()
+            s = torch.matmul(q, k.transpose(-2, -1)) * scale
+            s = s.masked_fill(~mask, float("-inf"))
+            p = torch.softmax(s.float(), dim=-1).to(q.dtype)
+            if self.dropout > 0:
+                p = nn.functional.dropout(p, self.dropout, self.training)
+            y = torch.matmul(p, v)
+        else:
+            raise ValueError(f"unknown attention backend {self.attention}")
+        y = y.transpose(1, 2).contiguous().view(B, T, E)
+        return self.c_proj(y)
+
+
+class MLP(nn.Module):
+    def __init__(self, config):
+        super().__init__()
+        self.c_fc = nn.Linear(config.n_embd, 4 * config.n_embd, bias=config.bias)
+        self.c_proj = nn.Linear(4 * config.n_embd, config.n_embd, bias=config.bias)
+
+    def forward(self, x):
+        return self.c_proj(ops.gelu(self.c_fc(x)))
+
+
+class Block(nn.Module):
+    def __init__(self, config):
+        super().__init__()
+        self.ln_1 = nn.LayerNorm(config.n_embd)
+        self.attn = CausalSelfAttention(config)
+        self.ln_2 = nn.LayerNorm(config.n_embd)
+        self.mlp = MLP(config)
+
+    def forward(self, x):
+        x = x + self.attn(self.ln_1(x))
+        x = x + self.mlp(self.ln_2(x))
+        return x
+
+
+class GPT2Model(nn.Module):
+    def __init__(self, config):
+        super().__init__()
+        self.config = config
+        self.transformer = nn.ModuleDict(dict(
+            wte=nn.Embedding(config.vocab_size, config.n_embd),
+            wpe=nn.Embedding(config.block_size, config.n_embd),
+            h=nn.ModuleList(Block(config) for _ in range(config.n_layer)),
+            ln_f=nn.LayerNorm(config.n_embd),
+        ))
+        self.lm_head = nn.Linear(config.n_embd, config.vocab_size, bias=False)
+        self.apply(self._init_weights)
+
+    def _init_weights(self, module):
+        if isinstance(module, nn.Linear):
+            nn.init.normal_(module.weight, mean=0.0, std=0.02)
+            if module.bias is not None:
+                nn.init.zeros_(module.bias)
+        elif isinstance(module, nn.Embedding):
+            nn.init.normal_(module.weight, mean=0.0, std=0.02)
+
+    def forward(self, idx, targets=None):
+        B, T = idx.shape
+        assert T <= self.config.block_size, (
+            f"sequence length {T} > block_size {self.config.block_size}"
+        )
+        pos = torch.arange(T, dtype=torch.long, device=idx.device)
+        tok = self.transformer.wte(idx)
+        posemb = self.transformer.wpe(pos)
+        x = tok + posemb
+        if self.config.dropout > 0:
+            x = nn.functional.dropout(x, self.config.dropout, self.training)
+        for block in self.transformer.h:
+            x = block(x)
+        x = self.transformer.ln_f(x)
+        logits = self.lm_head(x)
+        loss = None
+        if targets is not None:
+            loss = ops.cross_entropy(logits, targets)
+        return logits, loss
+
+    def num_params(self):
+        return sum(p.numel() for p in self.parameters())

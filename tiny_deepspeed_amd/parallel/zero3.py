@@ -1,0 +1,181 @@
+"""ZeRO-3 strategy: parameter sharding with just-in-time gathers.
+
+Builds the reference's *intent* (``/root/reference/tiny_deepspeed/core/zero/
+zero3/``) without its bugs: rank 0's parameters are
+broadcast like everyone else's, parameter memory on non-owners is actually
+released (0-numel storage + stream-ordered allocator for the per-layer
+gather buffers), biased Linears work, and the sync latch re-arms every
+iteration.
+
+Protocol per layer per iteration:
+  forward : broadcast full param from owner on the GATHER channel (own
+            stream + own RCCL communicator so it never serializes behind
+            grad reduces), compute, drop the gather buffer.
+  backward: re-gather for dW/dX, average-reduce dW to the owner on the
+            REDUCE channel, drop buffer and non-owner grads.
+  step    : owner updates its partition locally; NO step-time broadcast
+            (parameters are re-gathered JIT next iteration).
+
+True meta-init (the reference only plans on meta — SURVEY.md 2.11.9):
+wrap a meta-device model and each rank materializes ONLY the parameters it
+owns, sized for 288 GB HBM3E/GPU.
+"""
+
+import math
+
+import torch
+import torch.nn as nn
+
+from .. import modules as base
+from .. import ops
+from .. import optim as base_optim
+from ._grad import publish_grad, REDUCE_SHARD
+from ._zero_optim import _ZeroOptimMixin
+from .wrapper import ModelWrapper
+
+
+class _GatherMixin:
+    def _gather(self, param):
+        """Return the full tensor for `param` (broadcast from owner)."""
+        if param is None:
+            return None
+        comm = self._comm
+        if comm.world_size == 1:
+            return param
+        if comm.rank == param._tdsa_owner:
+            comm.gather_broadcast(param.data, src=param._tdsa_owner)
+            return param.data
+        buf = torch.empty(param._tdsa_full_shape, dtype=param.dtype,
+                          device=param.device)
+        comm.gather_broadcast(buf, src=param._tdsa_owner)
+        return buf
+
+
+class Linear(_GatherMixin, base.Linear):
+    def forward_callback(self, x, weight, bias):
+        w = self._gather(self.weight)
+        b = self._gather(self.bias)
+        self._comm.wait_gather()
+        return ops.linear_forward(x, w, b, tuner=self.tuner)
+        # gather buffers die here; the allocator reclaims them stream-safely
+
+    def backward_callback(self, dy, x):
+        w = self._gather(self.weight)
+        self._comm.wait_gather()
+        if self.weight.requires_grad:
+            dw = ops.linear_weight_grad(dy, x, tuner=self.tuner)
+            assert tuple(dw.shape) == self.weight._tdsa_full_shape
+            publish_grad(self._comm, self.weight, dw, REDUCE_SHARD)
+        if self.bias is not None and self.bias.requires_grad:
+            db = ops.linear_bias_grad(dy, tuner=self.tuner)
+            publish_grad(self._comm, self.bias, db, REDUCE_SHARD)
+        dx = ops.linear_input_grad(dy, w, tuner=self.tuner)
+        return dx, None, None
+
+
+class LayerNorm(_GatherMixin, base.LayerNorm):
+    def forward_callback(self, x, weight, bias):
+        w = self._gather(self.weight)
+        b = self._gather(self.bias)
+        self._comm.wait_gather()
+        return ops.layernorm_fwd(x, w, b, eps=self.eps, tuner=self.tuner)
+
+    def backward_callback(self, dy, x, mean, rstd):
+        w = self._gather(self.weight)
+        self._comm.wait_gather()
+        dx, ws = ops.layernorm_dx(dy, x, w, mean, rstd, tuner=self.tuner)
+        if self.weight.requires_grad:
+            dw, db = ops.layernorm_dwdb(ws, dtype=self.weight.dtype,
+                                        tuner=self.tuner)
+            publish_grad(self._comm, self.weight, dw, REDUCE_SHARD)
+            publish_grad(self._comm, self.bias, db, REDUCE_SHARD)
+        return dx, None, None
+
+
+class Embedding(_GatherMixin, base.Embedding):
+    def forward_callback(self, idx, weight):
+        w = self._gather(self.weight)
+        self._comm.wait_gather()
+        return ops.embedding_forward(w, idx, padding_idx=self.padding_idx,
+                                     tuner=self.tuner)
+
+    def backward_callback(self, dy, idx):
+        if self.weight.requires_grad:
+            dw = ops.embedding_weight_grad(idx, dy, self.num_embeddings,
+                                           padding_idx=self.padding_idx,
+                                           tuner=self.tuner)
+            publish_grad(self._comm, self.weight, dw, REDUCE_SHARD)
+        return None
+
+
+def _materialize(mod, pname, shape, dtype, device):
+    """Owner-local init for meta-wrapped models (module default inits)."""
+    t = torch.empty(shape, dtype=dtype, device=device)
+    if isinstance(mod, nn.Linear):
+        if pname == "weight":
+            nn.init.kaiming_uniform_(t, a=math.sqrt(5))
+        else:
+            fan_in = mod.in_features
+            bound = 1.0 / math.sqrt(fan_in) if fan_in > 0 else 0.0
+            nn.init.uniform_(t, -bound, bound)
+    elif isinstance(mod, nn.LayerNorm):
+        if pname == "weight":
+            nn.init.ones_(t)
+        else:
+            nn.init.zeros_(t)
+    elif isinstance(mod, nn.Embedding):
+        nn.init.normal_(t)
+    else:
+        nn.init.zeros_(t)
+    return t
+
+
+class Zero3(ModelWrapper):
+    swap_map = {
+        nn.Linear: Linear,
+        nn.LayerNorm: LayerNorm,
+        nn.Embedding: Embedding,
+    }
+
+    def __init__(self, module, parts, comm=None, device=None):
+        self._shard_device = device
+        super().__init__(module, parts=parts, comm=comm)
+
+    def _post_wrap(self):
+        rank = self.comm.rank
+        if self._shard_device is not None:
+            dev = torch.device(self._shard_device)
+        elif torch.cuda.is_available():
+            dev = torch.device("cuda", torch.cuda.current_device())
+        else:
+            dev = torch.device("cpu")
+        for mod in self.module.modules():
+            for pname, p in list(mod.named_parameters(recurse=False)):
+                if p._tdsa_owner == rank:
+                    if p.is_meta:
+                        p.data = _materialize(mod, pname, p._tdsa_full_shape,
+                                              p.dtype, dev)
+                else:
+                    # actually release non-owner parameter storage
+                    p.data = torch.empty(
+                        0, dtype=p.dtype,
+                        device=dev if p.is_meta else p.device,
+                    )
+
+
+class _Zero3OptimMixin(_ZeroOptimMixin):
+    broadcast_params_after_step = False  # params are gathered JIT instead
+
+
+class Zero3SGD(_Zero3OptimMixin, base_optim.SGD):
+    def __init__(self, parameters, param_part_table=None, ranks_map=None,
+                 comm=None, **kw):
+        self._setup_zero(param_part_table, ranks_map, comm)
+        super().__init__(parameters, **kw)
+
+
+class Zero3AdamW(_Zero3OptimMixin, base_optim.AdamW):
+    def __init__(self, parameters, param_part_table=None, ranks_map=None,
+                 comm=None, **kw):
+        self._setup_zero(param_part_table, ranks_map, comm)
+        super().__init__(parameters, **kw)
