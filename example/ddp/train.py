@@ -1,0 +1,39 @@
+"""DDP GPT-2 training (parity: /root/reference/example/ddp/train.py).
+
+Run: torchrun --standalone --local-addr 127.0.0.1 --nproc-per-node N example/ddp/train.py
+"""
+
+import os
+import sys
+
+sys.path.insert(0, os.path.abspath(os.path.join(os.path.dirname(__file__), "..", "..")))
+
+import torch
+import torch.distributed as dist
+
+from example.common import init_distributed, synthetic_batch
+from tiny_deepspeed_amd.models import GPTConfig, GPT2Model
+from tiny_deepspeed_amd import DDP, DDPAdamW
+
+rank, world_size, device = init_distributed()
+torch.manual_seed(0)  # same init on every rank (grads are averaged)
+dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+
+config = GPTConfig()
+model = GPT2Model(config).to(device=device, dtype=dtype)
+model = DDP(model)
+optimizer = DDPAdamW(model.named_parameters(), lr=1e-5, weight_decay=1e-1)
+
+# per-rank data shard: different seed per rank = data parallelism
+x, y = synthetic_batch(config.vocab_size, 1, config.block_size, device, seed=rank)
+
+for i in range(100):
+    model.require_backward_grad_sync = True
+    _, loss = model(x, y)
+    loss.backward()
+    optimizer.step()
+    loss = model.comm.all_reduce_scalar_avg(loss.detach())
+    if rank == 0:
+        print(f"iter {i} loss: {loss.item():.4f}")
+
+dist.destroy_process_group()

@@ -1,0 +1,182 @@
+"""Worker functions executed inside spawned gloo ranks (see dist_utils)."""
+
+from collections import OrderedDict
+
+import torch
+
+CFG = dict(block_size=32, vocab_size=96, n_layer=2, n_head=2, n_embd=32)
+ITERS = 4
+
+
+def make_cfg():
+    from tiny_deepspeed_amd.models import GPTConfig
+
+    return GPTConfig(**CFG)
+
+
+def batch(seed=0):
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randint(0, CFG["vocab_size"], (2, CFG["block_size"]), generator=g)
+    y = torch.randint(0, CFG["vocab_size"], (2, CFG["block_size"]), generator=g)
+    return x, y
+
+
+def single_device_losses():
+    import tiny_deepspeed_amd as tdsa
+    from tiny_deepspeed_amd.models import GPT2Model
+
+    torch.manual_seed(0)
+    model = tdsa.Single(GPT2Model(make_cfg()))
+    opt = tdsa.AdamW(model.named_parameters(), lr=1e-3, weight_decay=0.01)
+    x, y = batch()
+    losses = []
+    for _ in range(ITERS):
+        _, loss = model(x, y)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    return losses
+
+
+def train_strategy(rank, world, strategy):
+    import tiny_deepspeed_amd as tdsa
+    from tiny_deepspeed_amd.models import GPT2Model
+
+    torch.manual_seed(0)
+    model = GPT2Model(make_cfg())
+    ranks_map = ["cpu"] * world
+    x, y = batch()  # same data on all ranks -> parity with single device
+    if strategy == "ddp":
+        model = tdsa.DDP(model)
+        opt = tdsa.DDPAdamW(model.named_parameters(), lr=1e-3, weight_decay=0.01)
+    else:
+        with torch.device("meta"):
+            meta = GPT2Model(make_cfg())
+        parts, _ = tdsa.partition_tensors(
+            OrderedDict(meta.named_parameters()), ranks_map
+        )
+        wrap = {"zero1": tdsa.Zero1, "zero2": tdsa.Zero2, "zero3": tdsa.Zero3}[strategy]
+        optc = {"zero1": tdsa.Zero1AdamW, "zero2": tdsa.Zero2AdamW,
+                "zero3": tdsa.Zero3AdamW}[strategy]
+        model = wrap(model, parts)
+        opt = optc(model.named_parameters(), lr=1e-3, weight_decay=0.01,
+                   param_part_table=parts, ranks_map=ranks_map)
+    losses = []
+    for _ in range(ITERS):
+        model.require_backward_grad_sync = True
+        _, loss = model(x, y)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    return losses
+
+
+def ddp_grads_averaged(rank, world):
+    import tiny_deepspeed_amd as tdsa
+
+    torch.manual_seed(0)
+    lin = torch.nn.Linear(8, 8, bias=False)
+    model = tdsa.DDP(torch.nn.Sequential(lin))
+    torch.manual_seed(100 + rank)
+    x = torch.randn(4, 8)
+    model.require_backward_grad_sync = True
+    model(x).square().mean().backward()
+    model.comm.sync()
+    return lin.weight.grad.clone()
+
+
+def zero2_shard_check(rank, world):
+    import tiny_deepspeed_amd as tdsa
+    from tiny_deepspeed_amd.models import GPT2Model
+
+    torch.manual_seed(0)
+    model = GPT2Model(make_cfg())
+    with torch.device("meta"):
+        meta = GPT2Model(make_cfg())
+    parts, _ = tdsa.partition_tensors(
+        OrderedDict(meta.named_parameters()), ["cpu"] * world
+    )
+    model = tdsa.Zero2(model, parts)
+    x, y = batch()
+    model.require_backward_grad_sync = True
+    _, loss = model(x, y)
+    loss.backward()
+    model.comm.sync()
+    owned_with_grad, unowned_with_grad = 0, 0
+    for n, p in model.named_parameters():
+        if parts[n] == rank:
+            owned_with_grad += int(p.grad is not None)
+        else:
+            unowned_with_grad += int(p.grad is not None)
+    return owned_with_grad, unowned_with_grad
+
+
+def zero3_shard_check(rank, world):
+    import tiny_deepspeed_amd as tdsa
+    from tiny_deepspeed_amd.models import GPT2Model
+
+    torch.manual_seed(0)
+    model = GPT2Model(make_cfg())
+    with torch.device("meta"):
+        meta = GPT2Model(make_cfg())
+    parts, _ = tdsa.partition_tensors(
+        OrderedDict(meta.named_parameters()), ["cpu"] * world
+    )
+    model = tdsa.Zero3(model, parts)
+    owned_elems = sum(
+        p.numel() for n, p in model.named_parameters() if parts[n] == rank
+    )
+    unowned_elems = sum(
+        p.numel() for n, p in model.named_parameters() if parts[n] != rank
+    )
+    assert unowned_elems == 0  # params actually sharded
+    x, y = batch()
+    model.require_backward_grad_sync = True
+    _, loss = model(x, y)
+    loss.backward()
+    model.comm.sync()
+    return owned_elems, float(loss.item())
+
+
+def zero3_meta_init(rank, world):
+    import tiny_deepspeed_amd as tdsa
+    from tiny_deepspeed_amd.models import GPT2Model
+
+    with torch.device("meta"):
+        model = GPT2Model(make_cfg())
+        parts, _ = tdsa.partition_tensors(
+            OrderedDict(model.named_parameters()), ["cpu"] * world
+        )
+    model = tdsa.Zero3(model, parts, device="cpu")
+    opt = tdsa.Zero3AdamW(model.named_parameters(), lr=1e-3,
+                          param_part_table=parts, ranks_map=["cpu"] * world)
+    for n, p in model.named_parameters():
+        assert not p.is_meta
+        if parts[n] == rank:
+            assert p.numel() > 0
+        else:
+            assert p.numel() == 0
+    x, y = batch()
+    losses = []
+    for _ in range(3):
+        model.require_backward_grad_sync = True
+        _, loss = model(x, y)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    return losses
+
+
+def grad_accumulation(rank, world):
+    import tiny_deepspeed_amd as tdsa
+
+    torch.manual_seed(0)
+    lin = torch.nn.Linear(8, 8, bias=False)
+    model = tdsa.DDP(torch.nn.Sequential(lin))
+    torch.manual_seed(200 + rank)
+    xs = [torch.randn(4, 8) for _ in range(3)]
+    for i, x in enumerate(xs):
+        model.require_backward_grad_sync = i == len(xs) - 1
+        model(x).square().mean().backward()
+    model.comm.sync()
+    return lin.weight.grad.clone()

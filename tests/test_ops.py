@@ -1,0 +1,148 @@
+"""Op-level numerics: the op functions (torch fallback path on CPU) against
+plain torch references. The same suite runs the HIP kernels on GPU via
+tests/test_gpu_kernels.py."""
+
+import math
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from tiny_deepspeed_amd import ops
+
+
+def test_linear_ops_match_autograd():
+    torch.manual_seed(0)
+    x = torch.randn(4, 7, 16, requires_grad=True)
+    w = torch.randn(24, 16, requires_grad=True)
+    b = torch.randn(24, requires_grad=True)
+    y = ops.linear_forward(x, w, b)
+    ref = F.linear(x, w, b)
+    assert torch.allclose(y, ref, atol=1e-6)
+    dy = torch.randn_like(ref)
+    ref.backward(dy)
+    assert torch.allclose(ops.linear_input_grad(dy, w), x.grad, atol=1e-5)
+    assert torch.allclose(ops.linear_weight_grad(dy, x), w.grad, atol=1e-5)
+    assert torch.allclose(ops.linear_bias_grad(dy), b.grad, atol=1e-5)
+
+
+def test_layernorm_ops_match_autograd():
+    torch.manual_seed(0)
+    x = torch.randn(6, 5, 32, requires_grad=True)
+    w = torch.randn(32, requires_grad=True)
+    b = torch.randn(32, requires_grad=True)
+    y, mean, rstd = ops.layernorm_fwd(x, w, b)
+    ref = F.layer_norm(x, (32,), w, b)
+    assert torch.allclose(y, ref, atol=1e-5)
+    dy = torch.randn_like(ref)
+    ref.backward(dy)
+    dx, ws = ops.layernorm_dx(dy, x, w, mean, rstd)
+    dw, db = ops.layernorm_dwdb(ws)
+    assert torch.allclose(dx, x.grad, atol=1e-4)
+    assert torch.allclose(dw, w.grad, atol=1e-4)
+    assert torch.allclose(db, b.grad, atol=1e-4)
+
+
+def test_embedding_ops_match_autograd():
+    torch.manual_seed(0)
+    w = torch.randn(50, 16, requires_grad=True)
+    idx = torch.randint(0, 50, (3, 9))
+    y = ops.embedding_forward(w, idx)
+    ref = F.embedding(idx, w)
+    assert torch.allclose(y, ref)
+    dy = torch.randn_like(ref)
+    ref.backward(dy)
+    dw = ops.embedding_weight_grad(idx, dy, 50)
+    assert torch.allclose(dw, w.grad, atol=1e-5)
+
+
+def test_embedding_padding_idx():
+    w = torch.randn(10, 4)
+    idx = torch.tensor([[1, 2, 2, 3]])
+    dy = torch.ones(1, 4, 4)
+    dw = ops.embedding_weight_grad(idx, dy, 10, padding_idx=2)
+    assert dw[2].abs().sum() == 0
+    assert dw[1].abs().sum() > 0
+
+
+def test_gelu_matches_torch_tanh_gelu():
+    torch.manual_seed(0)
+    x = torch.randn(100, requires_grad=True)
+    y = ops.gelu(x)
+    ref = F.gelu(x, approximate="tanh")
+    assert torch.allclose(y, ref, atol=1e-6)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    g1 = x.grad.clone()
+    x.grad = None
+    ref.backward(dy)
+    assert torch.allclose(g1, x.grad, atol=1e-5)
+
+
+def test_cross_entropy_matches_torch():
+    torch.manual_seed(0)
+    logits = torch.randn(12, 33, requires_grad=True)
+    tgt = torch.randint(0, 33, (12,))
+    loss = ops.cross_entropy(logits, tgt)
+    ref_logits = logits.detach().clone().requires_grad_(True)
+    ref = F.cross_entropy(ref_logits, tgt)
+    assert torch.allclose(loss, ref, atol=1e-5)
+    loss.backward()
+    ref.backward()
+    assert torch.allclose(logits.grad, ref_logits.grad, atol=1e-5)
+
+
+def test_cross_entropy_ignore_index():
+    torch.manual_seed(0)
+    logits = torch.randn(8, 11, requires_grad=True)
+    tgt = torch.randint(0, 11, (8,))
+    tgt[2] = -100
+    tgt[5] = -100
+    loss = ops.cross_entropy(logits, tgt)
+    ref = F.cross_entropy(logits.detach(), tgt, ignore_index=-100)
+    assert torch.allclose(loss, ref, atol=1e-5)
+    loss.backward()
+    assert logits.grad[2].abs().sum() == 0
+
+
+def test_causal_attention_matches_math():
+    torch.manual_seed(0)
+    B, H, T, D = 2, 3, 16, 8
+    q = torch.randn(B, H, T, D, requires_grad=True)
+    k = torch.randn(B, H, T, D, requires_grad=True)
+    v = torch.randn(B, H, T, D, requires_grad=True)
+    out = ops.causal_attention(q, k, v)
+    scale = 1.0 / math.sqrt(D)
+    ref_q = q.detach().clone().requires_grad_(True)
+    ref_k = k.detach().clone().requires_grad_(True)
+    ref_v = v.detach().clone().requires_grad_(True)
+    ref = F.scaled_dot_product_attention(ref_q, ref_k, ref_v, is_causal=True,
+                                         scale=scale)
+    assert torch.allclose(out, ref, atol=1e-5)
+    do = torch.randn_like(out)
+    out.backward(do)
+    ref.backward(do)
+    assert torch.allclose(q.grad, ref_q.grad, atol=1e-4)
+    assert torch.allclose(k.grad, ref_k.grad, atol=1e-4)
+    assert torch.allclose(v.grad, ref_v.grad, atol=1e-4)
+
+
+def test_autotuner_picks_and_caches():
+    calls = {"a": 0, "b": 0}
+
+    def fa(x):
+        calls["a"] += 1
+        return x + 1
+
+    def fb(x):
+        calls["b"] += 1
+        return x + 1
+
+    tuner = ops.RuntimeAutoTuner(warmup=1, iters=2)
+    x = torch.randn(4)
+    out = tuner.choose("op", [fa, fb], x)
+    assert torch.allclose(out, x + 1)
+    before = dict(calls)
+    tuner.choose("op", [fa, fb], x)
+    # cached choice: exactly one more call total
+    assert calls["a"] + calls["b"] == before["a"] + before["b"] + 1

@@ -56,6 +56,7 @@ def swap_layers(model, swap_map, comm):
             # rebind the ORIGINAL parameter objects (no copy, no re-init)
             for pname, p in list(child.named_parameters(recurse=False)):
                 setattr(new_mod, pname, p)
+                p._tdsa_wrapped = True
             new_mod.training = child.training
             new_mod._comm = comm
             setattr(model, child_name, new_mod)
@@ -80,7 +81,7 @@ def check_all_params_wrapped(model):
     """Every parameter must live in a swapped module (reference parity:
     error_handling, wrapper.py:82-85)."""
     for name, p in model.named_parameters():
-        if not hasattr(p, "_tdsa_sync"):
+        if not getattr(p, "_tdsa_wrapped", False):
             raise RuntimeError(
                 f"parameter {name} belongs to an unsupported module type; "
                 "only Linear/LayerNorm/Embedding parameters are handled"

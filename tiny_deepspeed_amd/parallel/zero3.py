@@ -108,6 +108,17 @@ class Embedding(_GatherMixin, base.Embedding):
         return None
 
 
+def _rebind(mod, pname, old_param, new_data):
+    """Replace a meta Parameter with a real one (set_data cannot cross
+    device *types*), carrying the strategy flags over."""
+    new_p = nn.Parameter(new_data, requires_grad=old_param.requires_grad)
+    for attr in ("_tdsa_name", "_tdsa_sync", "_tdsa_full_shape",
+                 "_tdsa_owner", "_tdsa_wrapped"):
+        if hasattr(old_param, attr):
+            setattr(new_p, attr, getattr(old_param, attr))
+    setattr(mod, pname, new_p)
+
+
 def _materialize(mod, pname, shape, dtype, device):
     """Owner-local init for meta-wrapped models (module default inits)."""
     t = torch.empty(shape, dtype=dtype, device=device)
@@ -153,14 +164,17 @@ class Zero3(ModelWrapper):
             for pname, p in list(mod.named_parameters(recurse=False)):
                 if p._tdsa_owner == rank:
                     if p.is_meta:
-                        p.data = _materialize(mod, pname, p._tdsa_full_shape,
-                                              p.dtype, dev)
+                        _rebind(mod, pname, p,
+                                _materialize(mod, pname, p._tdsa_full_shape,
+                                             p.dtype, dev))
                 else:
                     # actually release non-owner parameter storage
-                    p.data = torch.empty(
-                        0, dtype=p.dtype,
-                        device=dev if p.is_meta else p.device,
-                    )
+                    empty = torch.empty(0, dtype=p.dtype,
+                                        device=dev if p.is_meta else p.device)
+                    if p.is_meta:
+                        _rebind(mod, pname, p, empty)
+                    else:
+                        p.data = empty
 
 
 class _Zero3OptimMixin(_ZeroOptimMixin):
