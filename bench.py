@@ -1,0 +1,176 @@
+"""Flagship benchmark: GPT-2 ZeRO-2 training throughput (tokens/sec, whole job).
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+For N > 1 the driver launches this under torch.distributed.run with one rank
+per GPU (RCCL over xGMI). Measures W untimed warmup steps, then exactly K
+steps bracketed by barrier + torch.cuda.synchronize() on both sides, takes
+the MAX elapsed over ranks, and rank 0 prints ONE JSON line.
+
+Metric/config per BASELINE.json: tokens/sec (whole node), GPT-2 ZeRO-2,
+synthetic data, random-init weights, bf16 compute with fp32 accumulation and
+fp32 optimizer state. Weak scaling: per-GPU batch is fixed as N grows.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+from collections import OrderedDict
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch
+import torch.distributed as dist
+
+from tiny_deepspeed_amd.models import GPTConfig, GPT2Model
+from tiny_deepspeed_amd import (
+    Single, AdamW,
+    DDP, DDPAdamW,
+    Zero1, Zero1AdamW,
+    Zero2, Zero2AdamW,
+    Zero3, Zero3AdamW,
+    partition_tensors,
+)
+
+WRAPPERS = {
+    "single": (Single, AdamW),
+    "ddp": (DDP, DDPAdamW),
+    "zero1": (Zero1, Zero1AdamW),
+    "zero2": (Zero2, Zero2AdamW),
+    "zero3": (Zero3, Zero3AdamW),
+}
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--model", default="gpt2-medium")
+    p.add_argument("--batch", type=int, default=8, help="per-GPU micro batch")
+    p.add_argument("--seq", type=int, default=1024)
+    p.add_argument("--parallel", default="zero2",
+                   choices=list(WRAPPERS.keys()))
+    args = p.parse_args()
+
+    have_gpu = torch.cuda.is_available()
+    world_size = int(os.getenv("WORLD_SIZE", "1"))
+    rank = int(os.getenv("RANK", "0"))
+    local_rank = int(os.getenv("LOCAL_RANK", str(rank)))
+    distributed = world_size > 1
+    if distributed:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29531")
+        dist.init_process_group(backend="nccl" if have_gpu else "gloo",
+                                init_method="env://", world_size=world_size,
+                                rank=rank)
+    if have_gpu:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+        dtype = torch.bfloat16
+    else:
+        device = torch.device("cpu")
+        dtype = torch.float32
+        # CPU run is a plumbing check only (BASELINE.json config #1):
+        # shrink so it completes in seconds.
+        args.model = "gpt2-small"
+        args.batch = 1
+        args.seq = min(args.seq, 128)
+
+    model_name = args.model
+    config = GPTConfig.named(model_name, block_size=max(args.seq, 64))
+
+    torch.manual_seed(1234)
+    model = GPT2Model(config).to(device=device, dtype=dtype)
+
+    wrapper_cls, optim_cls = WRAPPERS[args.parallel]
+    if args.parallel in ("zero1", "zero2", "zero3"):
+        ranks_map = [f"cuda:{i}" if have_gpu else "cpu"
+                     for i in range(world_size)]
+        with torch.device("meta"):
+            parts, _ = partition_tensors(
+                OrderedDict(GPT2Model(config).named_parameters()),
+                ranks_map=ranks_map, evenness_priority=0, verbose=False,
+            )
+        wrapped = wrapper_cls(model, parts)
+        optimizer = optim_cls(wrapped.named_parameters(), lr=1e-5,
+                              weight_decay=0.1, param_part_table=parts,
+                              ranks_map=ranks_map)
+    elif args.parallel == "ddp":
+        wrapped = wrapper_cls(model)
+        optimizer = optim_cls(wrapped.named_parameters(), lr=1e-5,
+                              weight_decay=0.1)
+    else:
+        wrapped = wrapper_cls(model)
+        optimizer = optim_cls(wrapped.named_parameters(), lr=1e-5,
+                              weight_decay=0.1)
+
+    g = torch.Generator().manual_seed(4242 + rank)
+    x = torch.randint(0, config.vocab_size, (args.batch, args.seq),
+                      generator=g).to(device)
+    y = torch.randint(0, config.vocab_size, (args.batch, args.seq),
+                      generator=g).to(device)
+
+    def step():
+        wrapped.require_backward_grad_sync = True
+        _, loss = wrapped(x, y)
+        loss.backward()
+        optimizer.step()
+        return loss
+
+    for _ in range(args.warmup):
+        step()
+
+    if distributed:
+        dist.barrier()
+    if have_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if have_gpu:
+        torch.cuda.synchronize()
+    if distributed:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    if distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device
+                         if have_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world_size if have_gpu else args.gpus
+    tokens_per_step = args.batch * args.seq * world_size
+    value = tokens_per_step * args.steps / elapsed
+    if rank == 0:
+        out = {
+            "metric": "tokens/sec (whole node) GPT-2 ZeRO-2",
+            "value": round(value, 1),
+            "unit": "tokens/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": model_name,
+                "global_batch": args.batch * world_size,
+                "seq_len": args.seq,
+                "parallelism": f"{args.parallel}-dp{world_size}",
+            },
+        }
+        print(json.dumps(out))
+
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

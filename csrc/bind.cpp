@@ -1,0 +1,308 @@
+// Python bindings for the tiny_deepspeed_amd CDNA4 kernel extension (_C).
+//
+// Host-side glue only: shape/dtype checks, output allocation, stream lookup.
+// All device code lives in csrc/kernels/*.hip (pure HIP, no torch headers).
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include <hip/hip_runtime_api.h>
+
+// ---- extern "C" launcher prototypes (csrc/kernels/*.hip) ------------------
+extern "C" {
+hipError_t tdsa_ln_fwd(const void*, const void*, const void*, void*, float*,
+                       float*, int, int, float, int, hipStream_t);
+int tdsa_ln_bwd_dx_stripes(int M);
+hipError_t tdsa_ln_bwd_dx(const void*, const void*, const void*, const float*,
+                          const float*, void*, float*, float*, int, int, int,
+                          hipStream_t);
+hipError_t tdsa_ln_bwd_dwdb(const float*, const float*, float*, float*, int,
+                            int, hipStream_t);
+hipError_t tdsa_gelu_fwd(const void*, void*, long long, int, hipStream_t);
+hipError_t tdsa_gelu_bwd(const void*, const void*, void*, long long, int,
+                         hipStream_t);
+hipError_t tdsa_column_sum(const void*, float*, void*, int, int, int,
+                           hipStream_t);
+hipError_t tdsa_embedding_fwd(const void*, const long long*, void*, long long,
+                              int, int, hipStream_t);
+hipError_t tdsa_embedding_bwd(const void*, const long long*, float*, long long,
+                              int, long long, int, hipStream_t);
+hipError_t tdsa_ce_fwd(const void*, const long long*, float*, float*, int*,
+                       long long, int, long long, int, hipStream_t);
+hipError_t tdsa_ce_bwd(const void*, const long long*, const float*, void*,
+                       long long, int, float, long long, int, hipStream_t);
+hipError_t tdsa_adamw_step(void*, const void*, float*, float*, float*, float*,
+                           int, int, float, float, float, float, float,
+                           long long, long long, int, int, hipStream_t);
+hipError_t tdsa_sgd_step(void*, const void*, float*, float*, int, int, float,
+                         float, float, float, int, int, int, long long, int,
+                         int, hipStream_t);
+hipError_t tdsa_attn_fwd(const void*, const void*, const void*, void*, float*,
+                         long long, int, float, hipStream_t);
+hipError_t tdsa_attn_bwd(const void*, const void*, const void*, const void*,
+                         const float*, const void*, void*, void*, void*, float*,
+                         long long, int, float, hipStream_t);
+}
+
+namespace {
+
+void check_hip(hipError_t err, const char* what) {
+  TORCH_CHECK(err == hipSuccess, what, ": ", hipGetErrorString(err));
+}
+
+int dtype_flag(const at::Tensor& t) {
+  if (t.scalar_type() == at::kBFloat16) return 1;
+  if (t.scalar_type() == at::kFloat) return 0;
+  TORCH_CHECK(false, "expected float32 or bfloat16, got ", t.scalar_type());
+  return -1;
+}
+
+hipStream_t cur_stream() {
+  return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+}
+
+#define CHECK_IN(t)                                                     \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");                     \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+// ---- layernorm ------------------------------------------------------------
+std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
+                                      double eps) {
+  CHECK_IN(x); CHECK_IN(w); CHECK_IN(b);
+  const int N = x.size(-1);
+  const long long M = x.numel() / N;
+  auto y = at::empty_like(x);
+  auto f32 = x.options().dtype(at::kFloat);
+  auto mean = at::empty({M}, f32);
+  auto rstd = at::empty({M}, f32);
+  check_hip(tdsa_ln_fwd(x.data_ptr(), w.data_ptr(), b.data_ptr(), y.data_ptr(),
+                        mean.data_ptr<float>(), rstd.data_ptr<float>(), (int)M,
+                        N, (float)eps, dtype_flag(x), cur_stream()),
+            "layernorm_fwd");
+  // mean/rstd viewed to the row shape of x
+  auto row_sizes = x.sizes().vec();
+  row_sizes.pop_back();
+  return {y, mean.view(row_sizes), rstd.view(row_sizes)};
+}
+
+std::vector<at::Tensor> layernorm_bwd_dx(at::Tensor dy, at::Tensor x,
+                                         at::Tensor w, at::Tensor mean,
+                                         at::Tensor rstd, int64_t n_stripes) {
+  CHECK_IN(dy); CHECK_IN(x); CHECK_IN(w);
+  (void)n_stripes;  // stripe count is chosen device-side for full occupancy
+  const int N = x.size(-1);
+  const long long M = x.numel() / N;
+  const int G = tdsa_ln_bwd_dx_stripes((int)M);
+  auto dx = at::empty_like(x);
+  auto f32 = x.options().dtype(at::kFloat);
+  auto pdw = at::empty({G, N}, f32);
+  auto pdb = at::empty({G, N}, f32);
+  auto meanc = mean.contiguous();
+  auto rstdc = rstd.contiguous();
+  check_hip(tdsa_ln_bwd_dx(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                           meanc.data_ptr<float>(), rstdc.data_ptr<float>(),
+                           dx.data_ptr(), pdw.data_ptr<float>(),
+                           pdb.data_ptr<float>(), (int)M, N, dtype_flag(x),
+                           cur_stream()),
+            "layernorm_bwd_dx");
+  return {dx, pdw, pdb};
+}
+
+std::vector<at::Tensor> layernorm_bwd_dwdb(at::Tensor pdw, at::Tensor pdb) {
+  CHECK_IN(pdw); CHECK_IN(pdb);
+  const int G = pdw.size(0);
+  const int N = pdw.size(1);
+  auto dw = at::empty({N}, pdw.options());
+  auto db = at::empty({N}, pdb.options());
+  check_hip(tdsa_ln_bwd_dwdb(pdw.data_ptr<float>(), pdb.data_ptr<float>(),
+                             dw.data_ptr<float>(), db.data_ptr<float>(), G, N,
+                             cur_stream()),
+            "layernorm_bwd_dwdb");
+  return {dw, db};
+}
+
+// ---- elementwise ----------------------------------------------------------
+at::Tensor gelu_fwd(at::Tensor x) {
+  CHECK_IN(x);
+  auto y = at::empty_like(x);
+  check_hip(tdsa_gelu_fwd(x.data_ptr(), y.data_ptr(), x.numel(), dtype_flag(x),
+                          cur_stream()),
+            "gelu_fwd");
+  return y;
+}
+
+at::Tensor gelu_bwd(at::Tensor dy, at::Tensor x) {
+  CHECK_IN(dy); CHECK_IN(x);
+  auto dx = at::empty_like(x);
+  check_hip(tdsa_gelu_bwd(dy.data_ptr(), x.data_ptr(), dx.data_ptr(), x.numel(),
+                          dtype_flag(x), cur_stream()),
+            "gelu_bwd");
+  return dx;
+}
+
+at::Tensor column_sum(at::Tensor dy) {
+  CHECK_IN(dy);
+  TORCH_CHECK(dy.dim() == 2, "column_sum expects 2-D input");
+  const int M = dy.size(0);
+  const int N = dy.size(1);
+  auto out32 = at::zeros({N}, dy.options().dtype(at::kFloat));
+  auto out = at::empty({N}, dy.options());
+  check_hip(tdsa_column_sum(dy.data_ptr(), out32.data_ptr<float>(),
+                            out.data_ptr(), M, N, dtype_flag(dy), cur_stream()),
+            "column_sum");
+  return out;
+}
+
+// ---- embedding ------------------------------------------------------------
+at::Tensor embedding_fwd(at::Tensor weight, at::Tensor idx) {
+  CHECK_IN(weight); CHECK_IN(idx);
+  TORCH_CHECK(idx.scalar_type() == at::kLong, "idx must be int64");
+  const long long R = idx.numel();
+  const int D = weight.size(1);
+  auto out = at::empty({R, (long long)D}, weight.options());
+  check_hip(tdsa_embedding_fwd(weight.data_ptr(), (const long long*)idx.data_ptr<int64_t>(),
+                               out.data_ptr(), R, D, dtype_flag(weight),
+                               cur_stream()),
+            "embedding_fwd");
+  return out;
+}
+
+at::Tensor embedding_bwd(at::Tensor dy, at::Tensor idx, int64_t num_embeddings,
+                         int64_t padding_idx) {
+  CHECK_IN(dy); CHECK_IN(idx);
+  TORCH_CHECK(idx.scalar_type() == at::kLong, "idx must be int64");
+  const long long R = idx.numel();
+  const int D = dy.size(-1);
+  auto dw32 = at::zeros({num_embeddings, (long long)D},
+                        dy.options().dtype(at::kFloat));
+  check_hip(tdsa_embedding_bwd(dy.data_ptr(), (const long long*)idx.data_ptr<int64_t>(),
+                               dw32.data_ptr<float>(), R, D, padding_idx,
+                               dtype_flag(dy), cur_stream()),
+            "embedding_bwd");
+  return dw32;
+}
+
+// ---- cross entropy --------------------------------------------------------
+std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets,
+                                          int64_t ignore_index) {
+  CHECK_IN(logits); CHECK_IN(targets);
+  TORCH_CHECK(logits.dim() == 2, "logits must be 2-D");
+  TORCH_CHECK(targets.scalar_type() == at::kLong, "targets must be int64");
+  const long long R = logits.size(0);
+  const int V = logits.size(1);
+  auto f32 = logits.options().dtype(at::kFloat);
+  auto lse = at::empty({R}, f32);
+  auto loss_sum = at::zeros({}, f32);
+  auto n_valid32 = at::zeros({}, logits.options().dtype(at::kInt));
+  check_hip(tdsa_ce_fwd(logits.data_ptr(), (const long long*)targets.data_ptr<int64_t>(),
+                        lse.data_ptr<float>(), loss_sum.data_ptr<float>(),
+                        n_valid32.data_ptr<int>(), R, V, ignore_index,
+                        dtype_flag(logits), cur_stream()),
+            "cross_entropy_fwd");
+  return {loss_sum, lse, n_valid32.to(at::kLong)};
+}
+
+at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets,
+                             at::Tensor lse, double dloss, int64_t n_valid,
+                             int64_t ignore_index) {
+  CHECK_IN(logits); CHECK_IN(targets); CHECK_IN(lse);
+  const long long R = logits.size(0);
+  const int V = logits.size(1);
+  auto dlogits = at::empty_like(logits);
+  const float scale = (float)(dloss / (double)std::max<int64_t>(n_valid, 1));
+  check_hip(tdsa_ce_bwd(logits.data_ptr(), (const long long*)targets.data_ptr<int64_t>(),
+                        lse.data_ptr<float>(), dlogits.data_ptr(), R, V, scale,
+                        ignore_index, dtype_flag(logits), cur_stream()),
+            "cross_entropy_bwd");
+  return dlogits;
+}
+
+// ---- fused optimizers -----------------------------------------------------
+void adamw_step(at::Tensor param, at::Tensor grad, at::Tensor m, at::Tensor v,
+                at::Tensor master, at::Tensor vmax, bool has_master,
+                bool amsgrad, double lr, double b1, double b2, double eps,
+                double wd, int64_t step) {
+  CHECK_IN(param); CHECK_IN(grad); CHECK_IN(m); CHECK_IN(v);
+  check_hip(tdsa_adamw_step(param.data_ptr(), grad.data_ptr(),
+                            m.data_ptr<float>(), v.data_ptr<float>(),
+                            has_master ? master.data_ptr<float>() : nullptr,
+                            amsgrad ? vmax.data_ptr<float>() : nullptr,
+                            has_master, amsgrad, (float)lr, (float)b1,
+                            (float)b2, (float)eps, (float)wd, step,
+                            param.numel(), dtype_flag(param), dtype_flag(grad),
+                            cur_stream()),
+            "adamw_step");
+}
+
+void sgd_step(at::Tensor param, at::Tensor grad, at::Tensor buf,
+              at::Tensor master, bool has_buf, bool has_master, double lr,
+              double momentum, double dampening, double wd, bool nesterov,
+              bool maximize, bool first_step) {
+  CHECK_IN(param); CHECK_IN(grad);
+  check_hip(tdsa_sgd_step(param.data_ptr(), grad.data_ptr(),
+                          has_buf ? buf.data_ptr<float>() : nullptr,
+                          has_master ? master.data_ptr<float>() : nullptr,
+                          has_buf, has_master, (float)lr, (float)momentum,
+                          (float)dampening, (float)wd, nesterov, maximize,
+                          first_step, param.numel(), dtype_flag(param),
+                          dtype_flag(grad), cur_stream()),
+            "sgd_step");
+}
+
+// ---- attention ------------------------------------------------------------
+std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                      double scale) {
+  CHECK_IN(q); CHECK_IN(k); CHECK_IN(v);
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == 64,
+              "attention kernel requires (B,H,T,64)");
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attention kernel is bf16");
+  TORCH_CHECK(q.size(2) % 64 == 0, "attention kernel requires T % 64 == 0");
+  const long long BH = (long long)q.size(0) * q.size(1);
+  const int T = q.size(2);
+  auto o = at::empty_like(q);
+  auto lse = at::empty({q.size(0), q.size(1), q.size(2)},
+                       q.options().dtype(at::kFloat));
+  check_hip(tdsa_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                          o.data_ptr(), lse.data_ptr<float>(), BH, T,
+                          (float)scale, cur_stream()),
+            "attention_fwd");
+  return {o, lse};
+}
+
+std::vector<at::Tensor> attention_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                      at::Tensor o, at::Tensor lse,
+                                      at::Tensor dout, double scale) {
+  CHECK_IN(q); CHECK_IN(k); CHECK_IN(v); CHECK_IN(o); CHECK_IN(dout);
+  auto lsec = lse.contiguous();
+  const long long BH = (long long)q.size(0) * q.size(1);
+  const int T = q.size(2);
+  auto dq = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto delta = at::empty({BH * T}, q.options().dtype(at::kFloat));
+  check_hip(tdsa_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                          o.data_ptr(), lsec.data_ptr<float>(), dout.data_ptr(),
+                          dq.data_ptr(), dk.data_ptr(), dv.data_ptr(),
+                          delta.data_ptr<float>(), BH, T, (float)scale,
+                          cur_stream()),
+            "attention_bwd");
+  return {dq, dk, dv};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("layernorm_fwd", &layernorm_fwd);
+  mod.def("layernorm_bwd_dx", &layernorm_bwd_dx);
+  mod.def("layernorm_bwd_dwdb", &layernorm_bwd_dwdb);
+  mod.def("gelu_fwd", &gelu_fwd);
+  mod.def("gelu_bwd", &gelu_bwd);
+  mod.def("column_sum", &column_sum);
+  mod.def("embedding_fwd", &embedding_fwd);
+  mod.def("embedding_bwd", &embedding_bwd);
+  mod.def("cross_entropy_fwd", &cross_entropy_fwd);
+  mod.def("cross_entropy_bwd", &cross_entropy_bwd);
+  mod.def("adamw_step", &adamw_step);
+  mod.def("sgd_step", &sgd_step);
+  mod.def("attention_fwd", &attention_fwd);
+  mod.def("attention_bwd", &attention_bwd);
+}

@@ -1,0 +1,187 @@
+#include "hip/hip_runtime.h"
+// Elementwise / reduction CDNA4 kernels: GELU (tanh) fwd+bwd, column-sum.
+//
+// The reference leaves GELU to torch autograd (/root/reference/example/
+// model.py:94) and the linear bias grad to dY.sum(0) (ops/linear.py:70-75);
+// here both are single-pass HBM-bound kernels (bf16x8 vectorized loads,
+// Guideline 13; grid-stride with a capped grid, Guideline 11).
+#include "common.h"
+
+namespace {
+
+constexpr float GC0 = 0.7978845608028654f;  // sqrt(2/pi)
+constexpr float GC1 = 0.044715f;
+
+DEV_INLINE float gelu_f(float x) {
+  return 0.5f * x * (1.0f + tanhf(GC0 * (x + GC1 * x * x * x)));
+}
+DEV_INLINE float gelu_df(float x) {
+  float t = tanhf(GC0 * (x + GC1 * x * x * x));
+  float dt = (1.0f - t * t) * GC0 * (1.0f + 3.0f * GC1 * x * x);
+  return 0.5f * (1.0f + t) + 0.5f * x * dt;
+}
+
+__global__ void gelu_fwd_bf16(const bf16* __restrict__ x, bf16* __restrict__ y,
+                              long long n8) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = global_tid(); i < n8; i += stride) {
+    short8v v = load8(x + i * 8);
+    short8v o;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) o[k] = bf_pack(gelu_f(bf_elem(v, k)));
+    store8(y + i * 8, o);
+  }
+}
+__global__ void gelu_fwd_f32(const float* __restrict__ x, float* __restrict__ y,
+                             long long n4) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = global_tid(); i < n4; i += stride) {
+    float4v v = *reinterpret_cast<const float4v*>(x + i * 4);
+    float4v o;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) o[k] = gelu_f(v[k]);
+    *reinterpret_cast<float4v*>(y + i * 4) = o;
+  }
+}
+__global__ void gelu_tail_bf16(const bf16* x, bf16* y, long long start, long long n) {
+  long long i = start + global_tid();
+  if (i < n) y[i] = f2bf(gelu_f(bf2f(x[i])));
+}
+__global__ void gelu_tail_f32(const float* x, float* y, long long start, long long n) {
+  long long i = start + global_tid();
+  if (i < n) y[i] = gelu_f(x[i]);
+}
+
+__global__ void gelu_bwd_bf16(const bf16* __restrict__ dy, const bf16* __restrict__ x,
+                              bf16* __restrict__ dx, long long n8) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = global_tid(); i < n8; i += stride) {
+    short8v dv = load8(dy + i * 8);
+    short8v xv = load8(x + i * 8);
+    short8v o;
+#pragma unroll
+    for (int k = 0; k < 8; ++k)
+      o[k] = bf_pack(bf_elem(dv, k) * gelu_df(bf_elem(xv, k)));
+    store8(dx + i * 8, o);
+  }
+}
+__global__ void gelu_bwd_f32(const float* __restrict__ dy, const float* __restrict__ x,
+                             float* __restrict__ dx, long long n4) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = global_tid(); i < n4; i += stride) {
+    float4v dv = *reinterpret_cast<const float4v*>(dy + i * 4);
+    float4v xv = *reinterpret_cast<const float4v*>(x + i * 4);
+    float4v o;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) o[k] = dv[k] * gelu_df(xv[k]);
+    *reinterpret_cast<float4v*>(dx + i * 4) = o;
+  }
+}
+__global__ void gelu_bwd_tail_bf16(const bf16* dy, const bf16* x, bf16* dx,
+                                   long long start, long long n) {
+  long long i = start + global_tid();
+  if (i < n) dx[i] = f2bf(bf2f(dy[i]) * gelu_df(bf2f(x[i])));
+}
+__global__ void gelu_bwd_tail_f32(const float* dy, const float* x, float* dx,
+                                  long long start, long long n) {
+  long long i = start + global_tid();
+  if (i < n) dx[i] = dy[i] * gelu_df(x[i]);
+}
+
+// db[N] = sum over M rows of dy[M,N]. Row-chunked: grid.y blocks each reduce
+// a chunk of rows for one 256-column slab, then one fp32 atomicAdd per
+// (chunk, column) — per-block partials first, Guideline 12.
+template <typename T>
+__global__ void colsum_kernel(const T* __restrict__ dy, float* __restrict__ out32,
+                              int M, int N, int rows_per_chunk) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= N) return;
+  const int r0 = blockIdx.y * rows_per_chunk;
+  const int r1 = min(M, r0 + rows_per_chunk);
+  float acc = 0.f;
+  for (int r = r0; r < r1; ++r) acc += (float)dy[(long long)r * N + c];
+  atomicAdd(&out32[c], acc);
+}
+
+template <typename T>
+__global__ void cast_from_f32(const float* __restrict__ in, T* __restrict__ out,
+                              long long n) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = global_tid(); i < n; i += stride) out[i] = (T)in[i];
+}
+
+}  // namespace
+
+extern "C" {
+
+hipError_t tdsa_gelu_fwd(const void* x, void* y, long long n, int is_bf16,
+                         hipStream_t stream) {
+  const int block = 256;
+  if (is_bf16) {
+    long long n8 = n / 8;
+    if (n8)
+      hipLaunchKernelGGL(gelu_fwd_bf16, dim3(ln_grid(n8, block)), dim3(block), 0,
+                         stream, (const bf16*)x, (bf16*)y, n8);
+    if (n % 8)
+      hipLaunchKernelGGL(gelu_tail_bf16, dim3(1), dim3(8), 0, stream,
+                         (const bf16*)x, (bf16*)y, n8 * 8, n);
+  } else {
+    long long n4 = n / 4;
+    if (n4)
+      hipLaunchKernelGGL(gelu_fwd_f32, dim3(ln_grid(n4, block)), dim3(block), 0,
+                         stream, (const float*)x, (float*)y, n4);
+    if (n % 4)
+      hipLaunchKernelGGL(gelu_tail_f32, dim3(1), dim3(4), 0, stream,
+                         (const float*)x, (float*)y, n4 * 4, n);
+  }
+  return hipGetLastError();
+}
+
+hipError_t tdsa_gelu_bwd(const void* dy, const void* x, void* dx, long long n,
+                         int is_bf16, hipStream_t stream) {
+  const int block = 256;
+  if (is_bf16) {
+    long long n8 = n / 8;
+    if (n8)
+      hipLaunchKernelGGL(gelu_bwd_bf16, dim3(ln_grid(n8, block)), dim3(block), 0,
+                         stream, (const bf16*)dy, (const bf16*)x, (bf16*)dx, n8);
+    if (n % 8)
+      hipLaunchKernelGGL(gelu_bwd_tail_bf16, dim3(1), dim3(8), 0, stream,
+                         (const bf16*)dy, (const bf16*)x, (bf16*)dx, n8 * 8, n);
+  } else {
+    long long n4 = n / 4;
+    if (n4)
+      hipLaunchKernelGGL(gelu_bwd_f32, dim3(ln_grid(n4, block)), dim3(block), 0,
+                         stream, (const float*)dy, (const float*)x, (float*)dx, n4);
+    if (n % 4)
+      hipLaunchKernelGGL(gelu_bwd_tail_f32, dim3(1), dim3(4), 0, stream,
+                         (const float*)dy, (const float*)x, (float*)dx, n4 * 4, n);
+  }
+  return hipGetLastError();
+}
+
+// out32 must be zero-filled fp32[N]; out (same dtype as dy) receives the cast.
+hipError_t tdsa_column_sum(const void* dy, float* out32, void* out, int M, int N,
+                           int is_bf16, hipStream_t stream) {
+  const int block = 256;
+  // size row chunks so total blocks ~ fill the chip
+  int chunks = 1;
+  long long col_blocks = (N + block - 1) / block;
+  while (col_blocks * chunks < 1024 && chunks * 128 < M) chunks *= 2;
+  int rows_per_chunk = (M + chunks - 1) / chunks;
+  dim3 grid(col_blocks, chunks);
+  if (is_bf16) {
+    hipLaunchKernelGGL(colsum_kernel<bf16>, grid, dim3(block), 0, stream,
+                       (const bf16*)dy, out32, M, N, rows_per_chunk);
+    hipLaunchKernelGGL(cast_from_f32<bf16>, dim3(ln_grid(N, block)), dim3(block),
+                       0, stream, out32, (bf16*)out, (long long)N);
+  } else {
+    hipLaunchKernelGGL(colsum_kernel<float>, grid, dim3(block), 0, stream,
+                       (const float*)dy, out32, M, N, rows_per_chunk);
+    hipLaunchKernelGGL(cast_from_f32<float>, dim3(ln_grid(N, block)), dim3(block),
+                       0, stream, out32, (float*)out, (long long)N);
+  }
+  return hipGetLastError();
+}
+
+}  // extern "C"

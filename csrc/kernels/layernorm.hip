@@ -1,0 +1,232 @@
+// LayerNorm forward / backward CDNA4 kernels (last-dim, affine+bias).
+//
+// Replaces the reference's three Triton kernels
+// (/root/reference/tiny_deepspeed/core/module/ops/layernorm.py:158-298) with
+// an MI355X-native design: one workgroup per row (grid-strided), fp32
+// accumulation, bf16x8 / float4 vectorized global traffic, and a
+// conflict-free two-pass dw/db reduction through per-block fp32 stripe
+// buffers instead of the reference's spin-lock atomic_cas scheme
+// (SURVEY.md 2.10A).
+#include "common.h"
+
+namespace {
+
+template <typename T> struct VecTraits;
+template <> struct VecTraits<float> {
+  static constexpr int W = 4;
+  typedef float4v V;
+  static DEV_INLINE V load(const float* p) { return *reinterpret_cast<const V*>(p); }
+  static DEV_INLINE float get(V v, int i) { return v[i]; }
+  static DEV_INLINE void set(V& v, int i, float x) { v[i] = x; }
+  static DEV_INLINE void store(float* p, V v) { *reinterpret_cast<V*>(p) = v; }
+};
+template <> struct VecTraits<bf16> {
+  static constexpr int W = 8;
+  typedef short8v V;
+  static DEV_INLINE V load(const bf16* p) { return load8(p); }
+  static DEV_INLINE float get(V v, int i) { return bf_elem(v, i); }
+  static DEV_INLINE void set(V& v, int i, float x) { v[i] = bf_pack(x); }
+  static DEV_INLINE void store(bf16* p, V v) { store8(p, v); }
+};
+
+template <typename T>
+__global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                              const T* __restrict__ b, T* __restrict__ y,
+                              float* __restrict__ mean, float* __restrict__ rstd,
+                              int M, int N, float eps) {
+  using VT = VecTraits<T>;
+  constexpr int W = VT::W;
+  __shared__ float scratch[1024 / WAVE];
+  const int tid = threadIdx.x;
+  const int nth = blockDim.x;
+  for (int row = blockIdx.x; row < M; row += gridDim.x) {
+    const T* xr = x + (long long)row * N;
+    float s = 0.f, ss = 0.f;
+    for (int i = tid * W; i < N; i += nth * W) {
+      typename VT::V v = VT::load(xr + i);
+#pragma unroll
+      for (int k = 0; k < W; ++k) {
+        float f = VT::get(v, k);
+        s += f;
+        ss += f * f;
+      }
+    }
+    s = block_sum(s, scratch);
+    ss = block_sum(ss, scratch);
+    const float mu = s / N;
+    const float var = fmaxf(ss / N - mu * mu, 0.0f);
+    const float rs = rsqrtf(var + eps);
+    if (tid == 0) {
+      mean[row] = mu;
+      rstd[row] = rs;
+    }
+    T* yr = y + (long long)row * N;
+    for (int i = tid * W; i < N; i += nth * W) {
+      typename VT::V xv = VT::load(xr + i);
+      typename VT::V wv = VT::load(w + i);
+      typename VT::V bv = VT::load(b + i);
+      typename VT::V ov;
+#pragma unroll
+      for (int k = 0; k < W; ++k) {
+        float xh = (VT::get(xv, k) - mu) * rs;
+        VT::set(ov, k, xh * VT::get(wv, k) + VT::get(bv, k));
+      }
+      VT::store(yr + i, ov);
+    }
+  }
+}
+
+// Backward dx + per-block dw/db stripe partials. Grid = G blocks; block g
+// handles rows g, g+G, ... and accumulates its dw/db into pdw[g*N..], so the
+// stripe write needs no atomics at all.
+template <typename T, int MAXITER>
+__global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__ x,
+                                 const T* __restrict__ w, const float* __restrict__ mean,
+                                 const float* __restrict__ rstd, T* __restrict__ dx,
+                                 float* __restrict__ pdw, float* __restrict__ pdb,
+                                 int M, int N) {
+  using VT = VecTraits<T>;
+  constexpr int W = VT::W;
+  __shared__ float scratch[1024 / WAVE];
+  const int tid = threadIdx.x;
+  const int nth = blockDim.x;
+  float accdw[MAXITER][W];
+  float accdb[MAXITER][W];
+#pragma unroll
+  for (int it = 0; it < MAXITER; ++it)
+#pragma unroll
+    for (int k = 0; k < W; ++k) accdw[it][k] = accdb[it][k] = 0.f;
+
+  for (int row = blockIdx.x; row < M; row += gridDim.x) {
+    const T* dyr = dy + (long long)row * N;
+    const T* xr = x + (long long)row * N;
+    const float mu = mean[row];
+    const float rs = rstd[row];
+    float c1 = 0.f, c2 = 0.f;
+    {
+      int it = 0;
+      for (int i = tid * W; i < N; i += nth * W, ++it) {
+        typename VT::V dv = VT::load(dyr + i);
+        typename VT::V xv = VT::load(xr + i);
+        typename VT::V wv = VT::load(w + i);
+#pragma unroll
+        for (int k = 0; k < W; ++k) {
+          float d = VT::get(dv, k);
+          float xh = (VT::get(xv, k) - mu) * rs;
+          float wdy = VT::get(wv, k) * d;
+          c1 += xh * wdy;
+          c2 += wdy;
+          accdw[it][k] += d * xh;
+          accdb[it][k] += d;
+        }
+      }
+    }
+    c1 = block_sum(c1, scratch) / N;
+    c2 = block_sum(c2, scratch) / N;
+    T* dxr = dx + (long long)row * N;
+    for (int i = tid * W; i < N; i += nth * W) {
+      typename VT::V dv = VT::load(dyr + i);
+      typename VT::V xv = VT::load(xr + i);
+      typename VT::V wv = VT::load(w + i);
+      typename VT::V ov;
+#pragma unroll
+      for (int k = 0; k < W; ++k) {
+        float d = VT::get(dv, k);
+        float xh = (VT::get(xv, k) - mu) * rs;
+        float wdy = VT::get(wv, k) * d;
+        VT::set(ov, k, (wdy - (xh * c1 + c2)) * rs);
+      }
+      VT::store(dxr + i, ov);
+    }
+  }
+  // stripe write: block-owned rows of pdw/pdb
+  float* sdw = pdw + (long long)blockIdx.x * N;
+  float* sdb = pdb + (long long)blockIdx.x * N;
+  {
+    int it = 0;
+    for (int i = tid * W; i < N; i += nth * W, ++it) {
+#pragma unroll
+      for (int k = 0; k < W; ++k) {
+        sdw[i + k] = accdw[it][k];
+        sdb[i + k] = accdb[it][k];
+      }
+    }
+  }
+}
+
+// Column-reduce the stripe buffers -> dw[N], db[N] (fp32 out; caller casts).
+__global__ void ln_bwd_dwdb_kernel(const float* __restrict__ pdw,
+                                   const float* __restrict__ pdb,
+                                   float* __restrict__ dw, float* __restrict__ db,
+                                   int G, int N) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= N) return;
+  float sw = 0.f, sb = 0.f;
+  for (int g = 0; g < G; ++g) {
+    sw += pdw[(long long)g * N + c];
+    sb += pdb[(long long)g * N + c];
+  }
+  dw[c] = sw;
+  db[c] = sb;
+}
+
+}  // namespace
+
+extern "C" {
+
+hipError_t tdsa_ln_fwd(const void* x, const void* w, const void* b, void* y,
+                       float* mean, float* rstd, int M, int N, float eps,
+                       int is_bf16, hipStream_t stream) {
+  const int block = 256;
+  const int grid = (M < 8 * 256) ? M : 8 * 256;
+  if (is_bf16) {
+    if (N % 8) return hipErrorInvalidValue;
+    hipLaunchKernelGGL(ln_fwd_kernel<bf16>, dim3(grid), dim3(block), 0, stream,
+                       (const bf16*)x, (const bf16*)w, (const bf16*)b, (bf16*)y,
+                       mean, rstd, M, N, eps);
+  } else {
+    if (N % 4) return hipErrorInvalidValue;
+    hipLaunchKernelGGL(ln_fwd_kernel<float>, dim3(grid), dim3(block), 0, stream,
+                       (const float*)x, (const float*)w, (const float*)b, (float*)y,
+                       mean, rstd, M, N, eps);
+  }
+  return hipGetLastError();
+}
+
+// G (stripe count) is chosen here and reported to the caller so it can size
+// pdw/pdb; call with pdw==nullptr to query G.
+int tdsa_ln_bwd_dx_stripes(int M) {
+  int g = M < 1024 ? M : 1024;
+  return g < 1 ? 1 : g;
+}
+
+hipError_t tdsa_ln_bwd_dx(const void* dy, const void* x, const void* w,
+                          const float* mean, const float* rstd, void* dx,
+                          float* pdw, float* pdb, int M, int N, int is_bf16,
+                          hipStream_t stream) {
+  const int block = 256;
+  const int grid = tdsa_ln_bwd_dx_stripes(M);
+  if (is_bf16) {
+    if (N % 8 || N > 2 * block * 8) return hipErrorInvalidValue;
+    hipLaunchKernelGGL((ln_bwd_dx_kernel<bf16, 2>), dim3(grid), dim3(block), 0,
+                       stream, (const bf16*)dy, (const bf16*)x, (const bf16*)w,
+                       mean, rstd, (bf16*)dx, pdw, pdb, M, N);
+  } else {
+    if (N % 4 || N > 2 * block * 4) return hipErrorInvalidValue;
+    hipLaunchKernelGGL((ln_bwd_dx_kernel<float, 2>), dim3(grid), dim3(block), 0,
+                       stream, (const float*)dy, (const float*)x, (const float*)w,
+                       mean, rstd, (float*)dx, pdw, pdb, M, N);
+  }
+  return hipGetLastError();
+}
+
+hipError_t tdsa_ln_bwd_dwdb(const float* pdw, const float* pdb, float* dw,
+                            float* db, int G, int N, hipStream_t stream) {
+  const int block = 256;
+  const int grid = (N + block - 1) / block;
+  hipLaunchKernelGGL(ln_bwd_dwdb_kernel, dim3(grid), dim3(block), 0, stream,
+                     pdw, pdb, dw, db, G, N);
+  return hipGetLastError();
+}
+
+}  // extern "C"

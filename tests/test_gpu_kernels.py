@@ -1,0 +1,241 @@
+"""GPU numerics: each CDNA4 HIP kernel against a plain PyTorch fp32 reference
+of the same op (SURVEY.md §4 test strategy, item (a)). All tests here require
+an MI355X and the in-tree _C extension."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from tiny_deepspeed_amd import ops
+from tiny_deepspeed_amd.ops import _ext
+
+
+def _close(out, ref, tol, name=""):
+    out = out.float()
+    ref = ref.float()
+    err = (out - ref).abs().max().item()
+    scale = max(ref.abs().max().item(), 1.0)
+    assert err <= tol * scale, f"{name}: max err {err} vs scale {scale} tol {tol}"
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _require_ext():
+    assert torch.cuda.is_available()
+    assert _ext.ext_available(), "HIP extension must be built (no eager fallback)"
+
+
+@pytest.mark.parametrize("dtype,tol", [(torch.float32, 1e-5), (torch.bfloat16, 2e-2)])
+def test_layernorm_gpu(dtype, tol):
+    torch.manual_seed(0)
+    M, N = 512, 768
+    x = torch.randn(M, N, device="cuda", dtype=dtype)
+    w = torch.randn(N, device="cuda", dtype=dtype)
+    b = torch.randn(N, device="cuda", dtype=dtype)
+    y, mean, rstd = ops.layernorm_fwd(x, w, b)
+    xf = x.float().requires_grad_(True)
+    wf = w.float().requires_grad_(True)
+    bf = b.float().requires_grad_(True)
+    ref = torch.nn.functional.layer_norm(xf, (N,), wf, bf)
+    _close(y, ref, tol, "ln fwd")
+    dy = torch.randn(M, N, device="cuda", dtype=dtype)
+    ref.backward(dy.float())
+    dx, ws = ops.layernorm_dx(dy, x, w, mean, rstd)
+    dw, db = ops.layernorm_dwdb(ws, dtype=dtype)
+    _close(dx, xf.grad, tol * 4, "ln dx")
+    _close(dw, wf.grad, tol * 4, "ln dw")
+    _close(db, bf.grad, tol * 4, "ln db")
+
+
+@pytest.mark.parametrize("dtype,tol", [(torch.float32, 1e-6), (torch.bfloat16, 1e-2)])
+def test_gelu_gpu(dtype, tol):
+    torch.manual_seed(0)
+    x = torch.randn(1000, 333, device="cuda", dtype=dtype)
+    y = ops.gelu_fwd(x)
+    xf = x.float()
+    ref = torch.nn.functional.gelu(xf, approximate="tanh")
+    _close(y, ref, tol, "gelu fwd")
+    dy = torch.randn_like(x)
+    dx = ops.gelu_bwd(dy, x)
+    xf = x.float().requires_grad_(True)
+    r2 = torch.nn.functional.gelu(xf, approximate="tanh")
+    r2.backward(dy.float())
+    _close(dx, xf.grad, tol, "gelu bwd")
+
+
+@pytest.mark.parametrize("dtype,tol", [(torch.float32, 1e-5), (torch.bfloat16, 1e-2)])
+def test_column_sum_gpu(dtype, tol):
+    torch.manual_seed(0)
+    dy = torch.randn(4096, 2304, device="cuda", dtype=dtype)
+    out = ops.linear_bias_grad(dy)
+    ref = dy.float().sum(dim=0)
+    _close(out, ref, tol, "column_sum")
+
+
+@pytest.mark.parametrize("dtype,tol", [(torch.float32, 1e-5), (torch.bfloat16, 1e-2)])
+def test_embedding_gpu(dtype, tol):
+    torch.manual_seed(0)
+    V, D = 50304, 768
+    w = torch.randn(V, D, device="cuda", dtype=dtype)
+    idx = torch.randint(0, V, (4, 1024), device="cuda")
+    y = ops.embedding_forward(w, idx)
+    ref = torch.nn.functional.embedding(idx, w.float())
+    _close(y, ref, tol, "emb fwd")
+    dy = torch.randn(4, 1024, D, device="cuda", dtype=dtype)
+    dw = ops.embedding_weight_grad(idx, dy, V)
+    dwr = torch.zeros(V, D, device="cuda", dtype=torch.float32)
+    dwr.index_add_(0, idx.reshape(-1), dy.reshape(-1, D).float())
+    _close(dw, dwr, tol, "emb bwd")
+
+
+def test_embedding_padding_idx_gpu():
+    w = torch.randn(10, 8, device="cuda", dtype=torch.float32)
+    idx = torch.tensor([[1, 2, 2, 3]], device="cuda")
+    dy = torch.ones(1, 4, 8, device="cuda")
+    dw = ops.embedding_weight_grad(idx, dy, 10, padding_idx=2)
+    assert dw[2].abs().sum().item() == 0
+    assert dw[1].abs().sum().item() > 0
+
+
+@pytest.mark.parametrize("dtype,tol", [(torch.float32, 1e-5), (torch.bfloat16, 2e-2)])
+def test_cross_entropy_gpu(dtype, tol):
+    torch.manual_seed(0)
+    R, V = 512, 50304
+    logits = torch.randn(R, V, device="cuda", dtype=dtype) * 3
+    tgt = torch.randint(0, V, (R,), device="cuda")
+    tgt[7] = -100
+    lf = logits.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(lf, tgt, ignore_index=-100)
+    logits.requires_grad_(True)
+    loss = ops.cross_entropy(logits, tgt)
+    _close(loss, ref, tol, "ce loss")
+    ref.backward()
+    loss.backward()
+    _close(logits.grad, lf.grad, tol, "ce grad")
+    assert logits.grad[7].abs().sum().item() == 0
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_adamw_step_gpu(dtype):
+    torch.manual_seed(0)
+    n = 12345
+    p = torch.randn(n, device="cuda", dtype=dtype)
+    g = torch.randn(n, device="cuda", dtype=dtype)
+    m = torch.randn(n, device="cuda").abs()
+    v = torch.randn(n, device="cuda").abs()
+    master = p.float().clone() if dtype != torch.float32 else None
+    # fp32 torch reference of the same math
+    pr = (master if master is not None else p).clone()
+    mr, vr = m.clone(), v.clone()
+    gr = g.float()
+    lr, b1, b2, eps, wd, step = 1e-3, 0.9, 0.999, 1e-8, 0.01, 3
+    pr.mul_(1 - lr * wd)
+    mr.mul_(b1).add_(gr, alpha=1 - b1)
+    vr.mul_(b2).addcmul_(gr, gr, value=1 - b2)
+    bc1, bc2 = 1 - b1 ** step, 1 - b2 ** step
+    pr.addcdiv_(mr, (vr / bc2).sqrt().add_(eps), value=-lr / bc1)
+    ops.adamw_step(p, g, m, v, master, step, lr, b1, b2, eps, wd)
+    _close(m, mr, 1e-6, "adamw m")
+    _close(v, vr, 1e-6, "adamw v")
+    if master is not None:
+        _close(master, pr, 1e-6, "adamw master")
+        _close(p, pr, 1e-2, "adamw p(bf16)")
+    else:
+        _close(p, pr, 1e-6, "adamw p")
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_sgd_step_gpu(dtype):
+    torch.manual_seed(0)
+    n = 9999
+    p = torch.randn(n, device="cuda", dtype=dtype)
+    g = torch.randn(n, device="cuda", dtype=dtype)
+    buf = torch.randn(n, device="cuda")
+    master = p.float().clone() if dtype != torch.float32 else None
+    pr = (master if master is not None else p).clone()
+    br = buf.clone()
+    gr = g.float()
+    lr, mom, damp, wd = 0.1, 0.9, 0.0, 0.01
+    gr = gr.add(pr, alpha=wd)
+    br.mul_(mom).add_(gr, alpha=1 - damp)
+    pr.add_(br, alpha=-lr)
+    ops.sgd_step(p, g, buf, master, lr, mom, damp, wd, False, False, False)
+    _close(buf, br, 1e-6, "sgd buf")
+    if master is not None:
+        _close(master, pr, 1e-6, "sgd master")
+    else:
+        _close(p, pr, 1e-6, "sgd p")
+
+
+def _attn_ref(q, k, v, scale):
+    qf, kf, vf = q.float(), k.float(), v.float()
+    T = q.shape[-2]
+    s = torch.matmul(qf, kf.transpose(-2, -1)) * scale
+    mask = torch.ones(T, T, dtype=torch.bool, device=q.device).tril()
+    s = s.masked_fill(~mask, float("-inf"))
+    return torch.softmax(s, dim=-1) @ vf
+
+
+def test_attention_fwd_gpu():
+    torch.manual_seed(0)
+    B, H, T, D = 2, 3, 256, 64
+    q = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16)
+    scale = 1.0 / math.sqrt(D)
+    o, lse = _ext.get_ext().attention_fwd(q, k, v, scale)
+    ref = _attn_ref(q, k, v, scale)
+    _close(o, ref, 2e-2, "attn fwd")
+    # lse check
+    s = torch.matmul(q.float(), k.float().transpose(-2, -1)) * scale
+    mask = torch.ones(T, T, dtype=torch.bool, device=q.device).tril()
+    s = s.masked_fill(~mask, float("-inf"))
+    lse_ref = torch.logsumexp(s, dim=-1)
+    _close(lse, lse_ref, 2e-2, "attn lse")
+
+
+def test_attention_bwd_gpu():
+    torch.manual_seed(1)
+    B, H, T, D = 2, 2, 256, 64
+    scale = 1.0 / math.sqrt(D)
+    q = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    out = ops.causal_attention(q, k, v)
+    dy = torch.randn_like(out)
+    out.backward(dy)
+    qf = q.detach().float().requires_grad_(True)
+    kf = k.detach().float().requires_grad_(True)
+    vf = v.detach().float().requires_grad_(True)
+    ref = _attn_ref(qf, kf, vf, scale)
+    ref.backward(dy.float())
+    _close(out, ref, 2e-2, "attn out")
+    _close(q.grad, qf.grad, 4e-2, "attn dq")
+    _close(k.grad, kf.grad, 4e-2, "attn dk")
+    _close(v.grad, vf.grad, 4e-2, "attn dv")
+
+
+def test_model_loss_decreases_gpu():
+    from tiny_deepspeed_amd import Single, AdamW
+    from tiny_deepspeed_amd.models import GPTConfig, GPT2Model
+
+    torch.manual_seed(0)
+    config = GPTConfig(n_layer=2, n_head=4, n_embd=256, block_size=256,
+                       vocab_size=1024)
+    model = GPT2Model(config).to(device="cuda", dtype=torch.bfloat16)
+    wrapped = Single(model)
+    opt = AdamW(wrapped.named_parameters(), lr=3e-4, weight_decay=0.0)
+    x = torch.randint(0, 1024, (2, 256), device="cuda")
+    y = torch.randint(0, 1024, (2, 256), device="cuda")
+    losses = []
+    for _ in range(20):
+        _, loss = wrapped(x, y)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0] - 0.5, losses

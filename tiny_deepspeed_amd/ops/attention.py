@@ -48,10 +48,20 @@ def _composite_bwd(q, k, v, o, lse, do, scale):
     return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype)
 
 
+def _kernel_supported(q):
+    """The fused CDNA4 kernel covers the GPT-2 family shapes: bf16,
+    head_dim 64, T % 64 == 0 (csrc/kernels/attention.hip contract). Other
+    shapes use the composite GPU path (rocBLAS matmuls) — documented, not a
+    silent eager fallback of a supported shape."""
+    return (
+        q.dtype == torch.bfloat16 and q.shape[-1] == 64 and q.shape[-2] % 64 == 0
+    )
+
+
 class _CausalAttentionFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale):
-        if _ext.use_native(q):
+        if _ext.use_native(q) and _kernel_supported(q):
             o, lse = _ext.get_ext().attention_fwd(
                 q.contiguous(), k.contiguous(), v.contiguous(), scale
             )
@@ -64,7 +74,7 @@ class _CausalAttentionFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
-        if _ext.use_native(q):
+        if _ext.use_native(q) and _kernel_supported(q):
             dq, dk, dv = _ext.get_ext().attention_bwd(
                 q.contiguous(), k.contiguous(), v.contiguous(),
                 o.contiguous(), lse, do.contiguous(), ctx.scale,
