@@ -41,6 +41,8 @@ hipError_t tdsa_attn_fwd(const void*, const void*, const void*, void*, float*,
 hipError_t tdsa_attn_bwd(const void*, const void*, const void*, const void*,
                          const float*, const void*, void*, void*, void*, float*,
                          long long, int, float, hipStream_t);
+hipError_t tdsa_dbg_mfma(const void*, const void*, float*, int, hipStream_t);
+hipError_t tdsa_dbg_stage(const void*, void*, int, hipStream_t);
 }
 
 namespace {
@@ -111,8 +113,8 @@ std::vector<at::Tensor> layernorm_bwd_dwdb(at::Tensor pdw, at::Tensor pdb) {
   CHECK_IN(pdw); CHECK_IN(pdb);
   const int G = pdw.size(0);
   const int N = pdw.size(1);
-  auto dw = at::empty({N}, pdw.options());
-  auto db = at::empty({N}, pdb.options());
+  auto dw = at::zeros({N}, pdw.options());
+  auto db = at::zeros({N}, pdb.options());
   check_hip(tdsa_ln_bwd_dwdb(pdw.data_ptr<float>(), pdb.data_ptr<float>(),
                              dw.data_ptr<float>(), db.data_ptr<float>(), G, N,
                              cur_stream()),
@@ -288,9 +290,30 @@ std::vector<at::Tensor> attention_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   return {dq, dk, dv};
 }
 
+// ---- debug probes ---------------------------------------------------------
+at::Tensor dbg_mfma(at::Tensor A, at::Tensor B, int64_t variant) {
+  CHECK_IN(A); CHECK_IN(B);
+  auto D = at::zeros({16, 16}, A.options().dtype(at::kFloat));
+  check_hip(tdsa_dbg_mfma(A.data_ptr(), B.data_ptr(), D.data_ptr<float>(),
+                          (int)variant, cur_stream()),
+            "dbg_mfma");
+  return D;
+}
+
+at::Tensor dbg_stage(at::Tensor in, int64_t transposed) {
+  CHECK_IN(in);
+  auto out = at::zeros_like(in);
+  check_hip(tdsa_dbg_stage(in.data_ptr(), out.data_ptr(), (int)transposed,
+                           cur_stream()),
+            "dbg_stage");
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("dbg_mfma", &dbg_mfma);
+  mod.def("dbg_stage", &dbg_stage);
   mod.def("layernorm_fwd", &layernorm_fwd);
   mod.def("layernorm_bwd_dx", &layernorm_bwd_dx);
   mod.def("layernorm_bwd_dwdb", &layernorm_bwd_dwdb);

@@ -84,18 +84,39 @@ DEV_INLINE unsigned pack2(float lo, float hi) {
   return u.u;
 }
 
-// Build the A-fragment (rows = this lane's c, k = 8g+e over 32-k slice `ks`)
-// from per-lane C-layout values packed as words w0=(r0,r1), w1=(r2,r3) per
-// 16-wide tile t. Source lane for e<4 is 32*(g&1)+c, for e>=4 that +16;
-// source tile index is 2*ks + (g>>1). (Derivation in repo docs/kernels.md.)
-DEV_INLINE bfrag frag_from_cpacked(const unsigned pw[4][2], int ks, int g, int c) {
-  const int qt = 2 * ks + (g >> 1);
+// Redistribute per-lane C-layout values (packed bf16 words w0=(r0,r1),
+// w1=(r2,r3) per 16-wide tile) into MFMA A-fragments whose k axis runs over
+// the C tiles' row axis. Dest lane (g,c), slice ks, elem e needs the value
+// of tile qt = 2*ks + (g>>1) from source lane 32*(g&1) + (e>=4 ? 16 : 0) + c.
+//
+// __shfl evaluates its operand on the SOURCE lane, so the tile index inside
+// the shuffled expression must not depend on the destination's registers:
+// shuffle every tile's words once (uniform expressions), then select.
+struct CShuffled {
+  unsigned sh[4][2][2];  // [tile][word][half: src L0 / L0+16]
+};
+
+DEV_INLINE CShuffled cshuffle(const unsigned pw[4][2], int g, int c) {
   const int L0 = 32 * (g & 1) + c;
+  CShuffled s;
+#pragma unroll
+  for (int qt = 0; qt < 4; ++qt)
+#pragma unroll
+    for (int wd = 0; wd < 2; ++wd) {
+      s.sh[qt][wd][0] = __shfl(pw[qt][wd], L0, WAVE);
+      s.sh[qt][wd][1] = __shfl(pw[qt][wd], L0 + 16, WAVE);
+    }
+  return s;
+}
+
+template <int KS>
+DEV_INLINE bfrag frag_from_shuffled(const CShuffled& s, int g) {
+  const bool hi = (g >> 1) != 0;  // tile = 2*KS + (g>>1): select, no scratch
   union { bfrag f; unsigned w[4]; } r;
-  r.w[0] = __shfl(pw[qt][0], L0, WAVE);
-  r.w[1] = __shfl(pw[qt][1], L0, WAVE);
-  r.w[2] = __shfl(pw[qt][0], L0 + 16, WAVE);
-  r.w[3] = __shfl(pw[qt][1], L0 + 16, WAVE);
+  r.w[0] = hi ? s.sh[2 * KS + 1][0][0] : s.sh[2 * KS][0][0];
+  r.w[1] = hi ? s.sh[2 * KS + 1][1][0] : s.sh[2 * KS][1][0];
+  r.w[2] = hi ? s.sh[2 * KS + 1][0][1] : s.sh[2 * KS][0][1];
+  r.w[3] = hi ? s.sh[2 * KS + 1][1][1] : s.sh[2 * KS][1][1];
   return r.f;
 }
 
@@ -200,13 +221,16 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
       for (int r = 0; r < 4; ++r) o_acc[dt][r] *= alpha_row[r];
 
     // PV: O[qrow][d] += P[qrow][key] V[key][d]
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      bfrag pa = frag_from_cpacked(pw, ks, g, c);
+    const CShuffled psh = cshuffle(pw, g, c);
+    {
+      bfrag pa0 = frag_from_shuffled<0>(psh, g);
+      bfrag pa1 = frag_from_shuffled<1>(psh, g);
 #pragma unroll
       for (int dt = 0; dt < 4; ++dt) {
-        bfrag vb = frag_read(lds_vt, dt * 16, ks * 32, g, c);
-        o_acc[dt] = MFMA_BF16_16x16x32(pa, vb, o_acc[dt]);
+        bfrag vb0 = frag_read(lds_vt, dt * 16, 0, g, c);
+        bfrag vb1 = frag_read(lds_vt, dt * 16, 32, g, c);
+        o_acc[dt] = MFMA_BF16_16x16x32(pa0, vb0, o_acc[dt]);
+        o_acc[dt] = MFMA_BF16_16x16x32(pa1, vb1, o_acc[dt]);
       }
     }
   }
@@ -332,16 +356,23 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
     }
 
     // dV[key][d] += P^T dO ; dK[key][d] += dS^T Q
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      bfrag ap = frag_from_cpacked(pwp, ks, g, c);
-      bfrag as = frag_from_cpacked(pws, ks, g, c);
+    const CShuffled pshp = cshuffle(pwp, g, c);
+    const CShuffled pshs = cshuffle(pws, g, c);
+    {
+      bfrag ap0 = frag_from_shuffled<0>(pshp, g);
+      bfrag ap1 = frag_from_shuffled<1>(pshp, g);
+      bfrag as0 = frag_from_shuffled<0>(pshs, g);
+      bfrag as1 = frag_from_shuffled<1>(pshs, g);
 #pragma unroll
       for (int dt = 0; dt < 4; ++dt) {
-        bfrag bdo = frag_read(lds_dot, dt * 16, ks * 32, g, c);
-        bfrag bq = frag_read(lds_qt, dt * 16, ks * 32, g, c);
-        dv_acc[dt] = MFMA_BF16_16x16x32(ap, bdo, dv_acc[dt]);
-        dk_acc[dt] = MFMA_BF16_16x16x32(as, bq, dk_acc[dt]);
+        bfrag bdo0 = frag_read(lds_dot, dt * 16, 0, g, c);
+        bfrag bdo1 = frag_read(lds_dot, dt * 16, 32, g, c);
+        bfrag bq0 = frag_read(lds_qt, dt * 16, 0, g, c);
+        bfrag bq1 = frag_read(lds_qt, dt * 16, 32, g, c);
+        dv_acc[dt] = MFMA_BF16_16x16x32(ap0, bdo0, dv_acc[dt]);
+        dv_acc[dt] = MFMA_BF16_16x16x32(ap1, bdo1, dv_acc[dt]);
+        dk_acc[dt] = MFMA_BF16_16x16x32(as0, bq0, dk_acc[dt]);
+        dk_acc[dt] = MFMA_BF16_16x16x32(as1, bq1, dk_acc[dt]);
       }
     }
   }
@@ -433,13 +464,16 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
     }
 
     // dQ[qrow][d] += dS K
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      bfrag as = frag_from_cpacked(pws, ks, g, c);
+    const CShuffled pshs = cshuffle(pws, g, c);
+    {
+      bfrag as0 = frag_from_shuffled<0>(pshs, g);
+      bfrag as1 = frag_from_shuffled<1>(pshs, g);
 #pragma unroll
       for (int dt = 0; dt < 4; ++dt) {
-        bfrag bk = frag_read(lds_kt, dt * 16, ks * 32, g, c);
-        dq_acc[dt] = MFMA_BF16_16x16x32(as, bk, dq_acc[dt]);
+        bfrag bk0 = frag_read(lds_kt, dt * 16, 0, g, c);
+        bfrag bk1 = frag_read(lds_kt, dt * 16, 32, g, c);
+        dq_acc[dt] = MFMA_BF16_16x16x32(as0, bk0, dq_acc[dt]);
+        dq_acc[dt] = MFMA_BF16_16x16x32(as1, bk1, dq_acc[dt]);
       }
     }
   }

@@ -156,19 +156,25 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__
 }
 
 // Column-reduce the stripe buffers -> dw[N], db[N] (fp32 out; caller casts).
+// 2-D grid: x covers columns, y covers stripe chunks; one fp32 atomicAdd per
+// (chunk, column) into the zero-initialized outputs (Guideline 12: per-block
+// partials first). A 1-D thread-per-column grid is N/256 workgroups — far
+// under the 256 CUs — and was the top kernel in the first bench profile.
 __global__ void ln_bwd_dwdb_kernel(const float* __restrict__ pdw,
                                    const float* __restrict__ pdb,
                                    float* __restrict__ dw, float* __restrict__ db,
-                                   int G, int N) {
+                                   int G, int N, int chunk) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= N) return;
+  const int g0 = blockIdx.y * chunk;
+  const int g1 = min(G, g0 + chunk);
   float sw = 0.f, sb = 0.f;
-  for (int g = 0; g < G; ++g) {
+  for (int g = g0; g < g1; ++g) {
     sw += pdw[(long long)g * N + c];
     sb += pdb[(long long)g * N + c];
   }
-  dw[c] = sw;
-  db[c] = sb;
+  atomicAdd(&dw[c], sw);
+  atomicAdd(&db[c], sb);
 }
 
 }  // namespace
@@ -221,12 +227,16 @@ hipError_t tdsa_ln_bwd_dx(const void* dy, const void* x, const void* w,
   return hipGetLastError();
 }
 
+// dw/db must be zero-initialized by the caller.
 hipError_t tdsa_ln_bwd_dwdb(const float* pdw, const float* pdb, float* dw,
                             float* db, int G, int N, hipStream_t stream) {
   const int block = 256;
-  const int grid = (N + block - 1) / block;
-  hipLaunchKernelGGL(ln_bwd_dwdb_kernel, dim3(grid), dim3(block), 0, stream,
-                     pdw, pdb, dw, db, G, N);
+  const int gx = (N + block - 1) / block;
+  int gy = 1;
+  while (gx * gy < 1024 && gy * 8 < G) gy *= 2;
+  const int chunk = (G + gy - 1) / gy;
+  hipLaunchKernelGGL(ln_bwd_dwdb_kernel, dim3(gx, gy), dim3(block), 0, stream,
+                     pdw, pdb, dw, db, G, N, chunk);
   return hipGetLastError();
 }
 

@@ -1,0 +1,47 @@
+"""On-GPU probe: determine the v_mfma_f32_16x16x32_bf16 fragment layout and
+validate the attention LDS staging path. Run on an MI355X box; prints a
+verdict per hypothesis and dumps raw D on mismatch."""
+
+import os
+import sys
+
+sys.path.insert(0, os.path.abspath(os.path.join(os.path.dirname(__file__), "..")))
+
+import torch
+
+from tiny_deepspeed_amd import _C
+
+
+def main():
+    torch.manual_seed(0)
+    dev = "cuda"
+    # integer-valued bf16-exact data, asymmetric (transpose-detecting)
+    A = (torch.randint(-8, 8, (16, 32), device=dev).to(torch.bfloat16))
+    B = (torch.randint(-8, 8, (32, 16), device=dev).to(torch.bfloat16))
+    ref = A.float() @ B.float()
+    for variant in (0, 1):
+        D = _C.dbg_mfma(A, B, variant)
+        checks = {
+            "D==A@B": (D - ref).abs().max().item(),
+            "D==(A@B).T": (D - ref.t()).abs().max().item(),
+        }
+        print(f"variant {variant}: " + "  ".join(
+            f"{k}: {v:.4f}" for k, v in checks.items()))
+    # if nothing matched, dump for offline permutation analysis
+    D0 = _C.dbg_mfma(A, B, 0)
+    torch.save({"A": A.cpu(), "B": B.cpu(), "D0": D0.cpu(),
+                "D1": _C.dbg_mfma(A, B, 1).cpu()},
+               "gpurun_out/mfma_probe.pt")
+
+    # staging probe: transposed image through scalar swizzled writes
+    M = torch.randn(64, 64, device=dev, dtype=torch.bfloat16)
+    out_t = _C.dbg_stage(M, 1)
+    err_t = (out_t.float() - M.float().t()).abs().max().item()
+    out_r = _C.dbg_stage(M, 0)
+    err_r = (out_r.float() - M.float()).abs().max().item()
+    print(f"stage rowmajor roundtrip err: {err_r:.4f} (expected 0)")
+    print(f"stage transposed err vs M^T: {err_t:.4f} (expected 0)")
+
+
+if __name__ == "__main__":
+    main()
