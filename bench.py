@@ -49,7 +49,7 @@ def main():
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--model", default="gpt2-medium")
-    p.add_argument("--batch", type=int, default=8, help="per-GPU micro batch")
+    p.add_argument("--batch", type=int, default=32, help="per-GPU micro batch")
     p.add_argument("--seq", type=int, default=1024)
     p.add_argument("--parallel", default="zero2",
                    choices=list(WRAPPERS.keys()))
