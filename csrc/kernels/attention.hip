@@ -2,22 +2,31 @@
 //
 // Replaces the reference's materialized (B,H,T,T) score path
 // (/root/reference/example/model.py:29-51) with an MI355X-native design:
-//   forward : per 64-row Q block, stream 64-key K/V tiles through
-//             XOR-swizzled LDS; S^T = mfma(K, Q) per 16x16 tile
+//   forward : per Q block (8 waves x 16 rows), stream 64-key K/V tiles
+//             through XOR-swizzled LDS; S^T = mfma(K, Q) per 16x16 tile
 //             (v_mfma_f32_16x16x32_bf16, fp32 accum) so the softmax row is
-//             lane-local; online softmax; P repacked to MFMA A-fragments
-//             with 8 ds_bpermute shuffles (no LDS round trip); O = P V with
-//             V staged transposed. Saves per-row logsumexp.
+//             lane-local; online softmax in exp2 space (log2e folded into
+//             the Q prescale); P repacked to MFMA A-fragments with
+//             tile-uniform ds_bpermute shuffles; O = P V with V staged
+//             transposed. Saves per-row logsumexp (natural log).
 //   backward: recompute-based two-kernel scheme (no atomics):
-//             dkv kernel owns a 64-key block and accumulates dK/dV over Q
-//             tiles; dq kernel owns a 64-row Q block. delta = rowsum(dO*O)
-//             precomputed by a small wave-reduction kernel.
+//             dkv kernel owns a 128-key block and accumulates dK/dV over Q
+//             tiles; dq kernel owns a 128-row Q block. delta = rowsum(dO*O)
+//             by a small wave-reduction kernel. The 1/sqrt(D) factor on
+//             dS is folded into the dK/dQ epilogue.
 //
-// Fragment layouts (gfx950, verified against rocm CK headers
-// ck_tile/ops/gemm/warp/warp_gemm_attribute_mfma_impl.hpp:195-221):
+// VALU-discipline (first profile showed 22:1 VALU:MFMA): every LDS offset
+// (fragment reads, staging stores) is computed once before the K/V loop;
+// causal masking runs only on diagonal tiles (wave-uniform branch); waves
+// whose rows lie entirely above/below a tile skip its compute.
+//
+// Fragment layouts (gfx950, verified on hardware by scripts/debug_mfma.py
+// and against ck_tile/ops/gemm/warp/warp_gemm_attribute_mfma_impl.hpp):
 //   A[16x32] : lane l, elem e(0..7) -> A[l%16][(l/16)*8 + e]
 //   B[32x16] : lane l, elem e      -> B[(l/16)*8 + e][l%16]
 //   C[16x16] : lane l, reg  r(0..3)-> C[(l/16)*4 + r][l%16]
+// (The hardware pairs A/B by register slot, so A and B must agree on the
+// slot->k map; the C map is fixed.)
 //
 // Contract: bf16 tensors (B,H,T,64) contiguous, T % 64 == 0. The Python op
 // (ops/attention.py) falls back to the composite path otherwise.
@@ -28,53 +37,16 @@
 
 namespace {
 
-constexpr int BLK = 64;   // q-rows / kv-keys per workgroup tile
+constexpr int KVB = 64;   // kv keys per LDS tile
 constexpr int D = 64;     // head dim (all GPT-2 sizes)
-constexpr int NW = 4;     // waves per workgroup
+constexpr float LOG2E = 1.4426950408889634f;
+constexpr float LN2 = 0.6931471805599453f;
 
-typedef __attribute__((ext_vector_type(8))) short bfrag;  // 8 bf16 (4 VGPRs)
+typedef __attribute__((ext_vector_type(8))) short bfrag;
 
-// XOR swizzle for a [64][64] bf16 LDS image with 128-byte rows: spreads the
-// 16-lane b128 read groups over 8 slots (Guideline 4: <=2-way).
+// XOR swizzle for a [64][64] bf16 LDS image with 128-byte rows (Guideline 4).
 DEV_INLINE int swz(int row, int byte_in_row) {
   return row * 128 + (byte_in_row ^ ((row & 7) << 4));
-}
-
-// Stage a [64][64] bf16 tile row-major from global into LDS, swizzled.
-// 256 threads, fully coalesced global reads (16B per lane, linear).
-DEV_INLINE void stage_rowmajor(const bf16* __restrict__ g, char* lds) {
-#pragma unroll
-  for (int rep = 0; rep < 2; ++rep) {
-    int chunk = threadIdx.x + rep * 256;  // 512 chunks of 8 bf16
-    int row = chunk >> 3;
-    int byte = (chunk & 7) * 16;
-    short8v v = load8(g + chunk * 8);
-    *reinterpret_cast<short8v*>(lds + swz(row, byte)) = v;
-  }
-}
-
-// Stage transposed: LDS image [d_or_col][64] from a global [64][d] tile.
-// Scalar u16 LDS writes (one-time per tile, shared by 4 waves).
-DEV_INLINE void stage_transposed(const bf16* __restrict__ g, char* lds) {
-#pragma unroll
-  for (int rep = 0; rep < 2; ++rep) {
-    int chunk = threadIdx.x + rep * 256;
-    int row = chunk >> 3;            // source row (key/qrow)
-    int c0 = (chunk & 7) * 8;        // source col (d)
-    short8v v = load8(g + chunk * 8);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int trow = c0 + j;             // dest row = d
-      *reinterpret_cast<short*>(lds + swz(trow, row * 2)) = v[j];
-    }
-  }
-}
-
-// Read one A/B fragment (16B) from a swizzled [64][64] image:
-// lane (g=l/16, c=l%16) reads row `row16 + c`, 8 bf16 at col `col8 + 8*g`.
-DEV_INLINE bfrag frag_read(const char* lds, int row16, int col8, int g, int c) {
-  int row = row16 + c;
-  return *reinterpret_cast<const bfrag*>(lds + swz(row, (col8 + 8 * g) * 2));
 }
 
 DEV_INLINE unsigned pack2(float lo, float hi) {
@@ -84,16 +56,66 @@ DEV_INLINE unsigned pack2(float lo, float hi) {
   return u.u;
 }
 
-// Redistribute per-lane C-layout values (packed bf16 words w0=(r0,r1),
-// w1=(r2,r3) per 16-wide tile) into MFMA A-fragments whose k axis runs over
-// the C tiles' row axis. Dest lane (g,c), slice ks, elem e needs the value
-// of tile qt = 2*ks + (g>>1) from source lane 32*(g&1) + (e>=4 ? 16 : 0) + c.
-//
-// __shfl evaluates its operand on the SOURCE lane, so the tile index inside
-// the shuffled expression must not depend on the destination's registers:
-// shuffle every tile's words once (uniform expressions), then select.
+DEV_INLINE bfrag lds_read16(const char* lds, int byte_off) {
+  return *reinterpret_cast<const bfrag*>(lds + byte_off);
+}
+
+// Per-lane precomputed addressing for one [64][64] staging + fragment-read
+// pattern. Staging is 512 chunks of 8 bf16 over NT threads.
+struct TileAddr {
+  int stage_src[2];   // global element offsets of this thread's chunks
+  int stage_dst[2];   // swizzled LDS byte offsets (row-major image)
+};
+
+template <int NT>
+DEV_INLINE TileAddr tile_addr(int tid) {
+  TileAddr a;
+#pragma unroll
+  for (int rep = 0; rep < 512 / (NT / 8) / 8; ++rep) {
+    int chunk = tid + rep * NT;
+    a.stage_src[rep] = chunk * 8;
+    a.stage_dst[rep] = swz(chunk >> 3, (chunk & 7) * 16);
+  }
+  return a;
+}
+
+template <int NT>
+DEV_INLINE void stage_rowmajor(const bf16* __restrict__ g, char* lds,
+                               const TileAddr& a) {
+#pragma unroll
+  for (int rep = 0; rep < 512 * 8 / NT / 8; ++rep)
+    *reinterpret_cast<short8v*>(lds + a.stage_dst[rep]) =
+        load8(g + a.stage_src[rep]);
+}
+
+// Transposed staging: dest row = source col. Scalar u16 writes.
+template <int NT>
+DEV_INLINE void stage_transposed(const bf16* __restrict__ g, char* lds,
+                                 const TileAddr& a, const int (&tdst)[2][8]) {
+#pragma unroll
+  for (int rep = 0; rep < 512 * 8 / NT / 8; ++rep) {
+    short8v v = load8(g + a.stage_src[rep]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      *reinterpret_cast<short*>(lds + tdst[rep][j]) = v[j];
+  }
+}
+
+template <int NT>
+DEV_INLINE void transposed_dst(int tid, int (&tdst)[2][8]) {
+#pragma unroll
+  for (int rep = 0; rep < 512 * 8 / NT / 8; ++rep) {
+    int chunk = tid + rep * NT;
+    int row = chunk >> 3;
+    int c0 = (chunk & 7) * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) tdst[rep][j] = swz(c0 + j, row * 2);
+  }
+}
+
+// C-layout -> A-fragment redistribution (see header comment of cshuffle).
 struct CShuffled {
-  unsigned sh[4][2][2];  // [tile][word][half: src L0 / L0+16]
+  unsigned sh[4][2][2];  // [tile][word][half]
 };
 
 DEV_INLINE CShuffled cshuffle(const unsigned pw[4][2], int g, int c) {
@@ -111,7 +133,7 @@ DEV_INLINE CShuffled cshuffle(const unsigned pw[4][2], int g, int c) {
 
 template <int KS>
 DEV_INLINE bfrag frag_from_shuffled(const CShuffled& s, int g) {
-  const bool hi = (g >> 1) != 0;  // tile = 2*KS + (g>>1): select, no scratch
+  const bool hi = (g >> 1) != 0;
   union { bfrag f; unsigned w[4]; } r;
   r.w[0] = hi ? s.sh[2 * KS + 1][0][0] : s.sh[2 * KS][0][0];
   r.w[1] = hi ? s.sh[2 * KS + 1][1][0] : s.sh[2 * KS][1][0];
@@ -120,88 +142,126 @@ DEV_INLINE bfrag frag_from_shuffled(const CShuffled& s, int g) {
   return r.f;
 }
 
+// load a global B/A-style fragment (row r0+c, 8 elems at col c0+8g) scaled
+DEV_INLINE bfrag load_frag_scaled(const bf16* p, int row, int col, float s) {
+  short8v v = load8(p + row * D + col);
+  union { bfrag f; short w[8]; } r;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) r.w[e] = bf_pack(bf_elem(v, e) * s);
+  return r.f;
+}
+
+DEV_INLINE bfrag load_frag(const bf16* p, int row, int col) {
+  union { bfrag f; short8v v; } r;
+  r.v = load8(p + row * D + col);
+  return r.f;
+}
+
 // ---------------------------------------------------------------------------
 // Forward
 // ---------------------------------------------------------------------------
-__launch_bounds__(256, 2)
+template <int NW>
+__launch_bounds__(NW * WAVE, 2)
 __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
                                 const bf16* __restrict__ k,
                                 const bf16* __restrict__ v,
                                 bf16* __restrict__ o, float* __restrict__ lse,
                                 int T, float scale) {
-  __shared__ __attribute__((aligned(16))) char smem[2 * BLK * D * 2];
+  constexpr int BM = NW * 16;  // q rows per workgroup
+  constexpr int NT = NW * WAVE;
+  __shared__ __attribute__((aligned(16))) char smem[2 * KVB * D * 2];
   char* lds_k = smem;                 // [64][64] keys row-major
-  char* lds_vt = smem + BLK * D * 2;  // [64(d)][64(key)] V transposed
+  char* lds_vt = smem + KVB * D * 2;  // [64(d)][64(key)] V transposed
 
-  const int qb = blockIdx.x;          // q block
-  const long long bh = blockIdx.y;    // batch*head
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int w = threadIdx.x / WAVE;   // wave id: owns q rows w*16..w*16+15
+  const int qb = blockIdx.x;
+  const long long bh = blockIdx.y;
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid / WAVE;
   const int g = lane >> 4;
   const int c = lane & 15;
 
-  const bf16* qp = q + (bh * T + qb * BLK) * D;
+  const bf16* qp = q + (bh * T + qb * BM) * D;
   const bf16* kp = k + bh * T * D;
   const bf16* vp = v + bh * T * D;
 
-  // Q B-fragments for this wave (rows w*16+c, 2 d-slices), pre-scaled into
-  // the softmax instead (scale applied to S).
+  // addressing, hoisted out of the K/V loop
+  const TileAddr ta = tile_addr<NT>(tid);
+  int tdst[2][8];
+  transposed_dst<NT>(tid, tdst);
+  int kf_off[4][2], vf_off[4][2];
+#pragma unroll
+  for (int t16 = 0; t16 < 4; ++t16)
+#pragma unroll
+    for (int ds = 0; ds < 2; ++ds) {
+      kf_off[t16][ds] = swz(t16 * 16 + c, (ds * 32 + 8 * g) * 2);
+      vf_off[t16][ds] = kf_off[t16][ds];  // same pattern on the V^T image
+    }
+
+  // Q fragments, prescaled by scale*log2e (softmax runs in exp2 space)
+  const float qs = scale * LOG2E;
   bfrag q_frag[2];
 #pragma unroll
   for (int ds = 0; ds < 2; ++ds)
-    q_frag[ds] = *reinterpret_cast<const bfrag*>(
-        qp + (w * 16 + c) * D + ds * 32 + 8 * g);
+    q_frag[ds] = load_frag_scaled(qp, w * 16 + c, ds * 32 + 8 * g, qs);
 
   f32x4 o_acc[4] = {};
   float m_run = -INFINITY;
   float l_run = 0.f;
 
-  const int n_kv = qb + 1;  // causal: kv tiles 0..qb
+  const int row_lo = qb * BM + w * 16;       // this wave's first q row
+  const int row_me = row_lo + c;             // this lane's q row
+  const int n_kv = (qb + 1) * BM / KVB;
   for (int j = 0; j < n_kv; ++j) {
     __syncthreads();
-    stage_rowmajor(kp + j * BLK * D, lds_k);
-    stage_transposed(vp + j * BLK * D, lds_vt);
+    stage_rowmajor<NT>(kp + j * KVB * D, lds_k, ta);
+    stage_transposed<NT>(vp + j * KVB * D, lds_vt, ta, tdst);
     __syncthreads();
 
-    // S^T tiles: C[key = 16*sub + 4g + r][qrow = c]
+    const int key0 = j * KVB;
+    if (key0 > row_lo + 15) continue;        // wave fully above this tile
+
+    // S^T tiles: C[key = 16*sub + 4g + r][qrow = c]  (values are log2-scaled)
     f32x4 st[4];
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub) {
       f32x4 acc = {};
-#pragma unroll
-      for (int ds = 0; ds < 2; ++ds) {
-        bfrag a = frag_read(lds_k, sub * 16, ds * 32, g, c);
-        acc = MFMA_BF16_16x16x32(a, q_frag[ds], acc);
-      }
+      acc = MFMA_BF16_16x16x32(lds_read16(lds_k, kf_off[sub][0]), q_frag[0], acc);
+      acc = MFMA_BF16_16x16x32(lds_read16(lds_k, kf_off[sub][1]), q_frag[1], acc);
       st[sub] = acc;
     }
 
-    // scale + causal mask + online softmax (row = this lane's qrow)
-    const int qrow = qb * BLK + w * 16 + c;
+    const bool diag = key0 + KVB - 1 > row_lo;  // some key may exceed a row
     float mt = -INFINITY;
+    if (diag) {
 #pragma unroll
-    for (int sub = 0; sub < 4; ++sub)
+      for (int sub = 0; sub < 4; ++sub)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int key = j * BLK + sub * 16 + 4 * g + r;
-        float s = st[sub][r] * scale;
-        s = (key <= qrow) ? s : -INFINITY;
-        st[sub][r] = s;
-        mt = fmaxf(mt, s);
-      }
+        for (int r = 0; r < 4; ++r) {
+          const int key = key0 + sub * 16 + 4 * g + r;
+          float s = (key <= row_me) ? st[sub][r] : -INFINITY;
+          st[sub][r] = s;
+          mt = fmaxf(mt, s);
+        }
+    } else {
+#pragma unroll
+      for (int sub = 0; sub < 4; ++sub)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) mt = fmaxf(mt, st[sub][r]);
+    }
     mt = fmaxf(mt, __shfl_xor(mt, 16, WAVE));
     mt = fmaxf(mt, __shfl_xor(mt, 32, WAVE));
     const float m_new = fmaxf(m_run, mt);
-    const float alpha = __expf(m_run - m_new);  // m_run=-inf,m_new=-inf can't happen (key 0 valid)
+    const float alpha = exp2f(m_run - m_new);
     float psum = 0.f;
     unsigned pw[4][2];
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub) {
-      float p0, p1, p2, p3;
-      p0 = (st[sub][0] == -INFINITY) ? 0.f : __expf(st[sub][0] - m_new);
-      p1 = (st[sub][1] == -INFINITY) ? 0.f : __expf(st[sub][1] - m_new);
-      p2 = (st[sub][2] == -INFINITY) ? 0.f : __expf(st[sub][2] - m_new);
-      p3 = (st[sub][3] == -INFINITY) ? 0.f : __expf(st[sub][3] - m_new);
+      // masked scores are -inf and m_new is finite, so exp2 gives exact 0
+      const float p0 = exp2f(st[sub][0] - m_new);
+      const float p1 = exp2f(st[sub][1] - m_new);
+      const float p2 = exp2f(st[sub][2] - m_new);
+      const float p3 = exp2f(st[sub][3] - m_new);
       psum += p0 + p1 + p2 + p3;
       pw[sub][0] = pack2(p0, p1);
       pw[sub][1] = pack2(p2, p3);
@@ -211,43 +271,43 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
     l_run = l_run * alpha + psum;
     m_run = m_new;
 
-    // O rescale: O C-layout rows are qrow = 4g+r -> fetch that row's alpha
-    float alpha_row[4];
+    // O rescale (skip when alpha == 1 for every row of the wave)
+    if (__any(alpha != 1.0f)) {
+      float alpha_row[4];
 #pragma unroll
-    for (int r = 0; r < 4; ++r) alpha_row[r] = __shfl(alpha, 4 * g + r, WAVE);
+      for (int r = 0; r < 4; ++r) alpha_row[r] = __shfl(alpha, 4 * g + r, WAVE);
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt)
+      for (int dt = 0; dt < 4; ++dt)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) o_acc[dt][r] *= alpha_row[r];
+        for (int r = 0; r < 4; ++r) o_acc[dt][r] *= alpha_row[r];
+    }
 
-    // PV: O[qrow][d] += P[qrow][key] V[key][d]
+    // PV
     const CShuffled psh = cshuffle(pw, g, c);
-    {
-      bfrag pa0 = frag_from_shuffled<0>(psh, g);
-      bfrag pa1 = frag_from_shuffled<1>(psh, g);
+    const bfrag pa0 = frag_from_shuffled<0>(psh, g);
+    const bfrag pa1 = frag_from_shuffled<1>(psh, g);
 #pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-        bfrag vb0 = frag_read(lds_vt, dt * 16, 0, g, c);
-        bfrag vb1 = frag_read(lds_vt, dt * 16, 32, g, c);
-        o_acc[dt] = MFMA_BF16_16x16x32(pa0, vb0, o_acc[dt]);
-        o_acc[dt] = MFMA_BF16_16x16x32(pa1, vb1, o_acc[dt]);
-      }
+    for (int dt = 0; dt < 4; ++dt) {
+      o_acc[dt] = MFMA_BF16_16x16x32(pa0, lds_read16(lds_vt, vf_off[dt][0]),
+                                     o_acc[dt]);
+      o_acc[dt] = MFMA_BF16_16x16x32(pa1, lds_read16(lds_vt, vf_off[dt][1]),
+                                     o_acc[dt]);
     }
   }
 
-  // epilogue: O /= l ; lse = m + log(l)
+  // epilogue: O /= l ; lse = (m + log2(l)) * ln2
   float linv_row[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r)
     linv_row[r] = 1.0f / __shfl(l_run, 4 * g + r, WAVE);
-  bf16* op = o + (bh * T + qb * BLK + w * 16) * D;
+  bf16* op = o + (bh * T + qb * BM + w * 16) * D;
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt)
 #pragma unroll
     for (int r = 0; r < 4; ++r)
       op[(4 * g + r) * D + dt * 16 + c] = f2bf(o_acc[dt][r] * linv_row[r]);
   if (lane < 16) {
-    lse[bh * T + qb * BLK + w * 16 + c] = m_run + __logf(l_run);
+    lse[bh * T + qb * BM + w * 16 + c] = (m_run + log2f(l_run)) * LN2;
   }
 }
 
@@ -267,9 +327,10 @@ __global__ void attn_delta_kernel(const bf16* __restrict__ dout,
 }
 
 // ---------------------------------------------------------------------------
-// Backward dK/dV: one workgroup per 64-key block; wave w owns keys w*16..+15.
+// Backward dK/dV: one workgroup per (NW*16)-key block; wave w owns 16 keys.
 // ---------------------------------------------------------------------------
-__launch_bounds__(256, 2)
+template <int NW>
+__launch_bounds__(NW * WAVE, 2)
 __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
                                     const bf16* __restrict__ k,
                                     const bf16* __restrict__ v,
@@ -278,19 +339,21 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
                                     const float* __restrict__ delta,
                                     bf16* __restrict__ dk, bf16* __restrict__ dv,
                                     int T, float scale) {
-  // carves: Q row-major, dO row-major, Q^T, dO^T, lse+delta tiles
-  __shared__ __attribute__((aligned(16))) char smem[4 * BLK * D * 2 + 2 * BLK * 4];
+  constexpr int BK = NW * 16;  // keys per workgroup
+  constexpr int NT = NW * WAVE;
+  __shared__ __attribute__((aligned(16))) char smem[4 * KVB * D * 2 + 2 * KVB * 4];
   char* lds_q = smem;
-  char* lds_do = smem + BLK * D * 2;
-  char* lds_qt = smem + 2 * BLK * D * 2;
-  char* lds_dot = smem + 3 * BLK * D * 2;
-  float* lds_lse = reinterpret_cast<float*>(smem + 4 * BLK * D * 2);
-  float* lds_dlt = lds_lse + BLK;
+  char* lds_do = smem + KVB * D * 2;
+  char* lds_qt = smem + 2 * KVB * D * 2;
+  char* lds_dot = smem + 3 * KVB * D * 2;
+  float* lds_lse = reinterpret_cast<float*>(smem + 4 * KVB * D * 2);
+  float* lds_dlt = lds_lse + KVB;
 
-  const int jb = blockIdx.x;        // kv block
+  const int jb = blockIdx.x;
   const long long bh = blockIdx.y;
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int w = threadIdx.x / WAVE;  // wave owns keys w*16..w*16+15
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid / WAVE;
   const int g = lane >> 4;
   const int c = lane & 15;
 
@@ -299,55 +362,69 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
   const bf16* vp = v + bh * T * D;
   const bf16* dop = dout + bh * T * D;
 
-  // This wave's K and V B-fragments (key = c, d = ds*32+8g+e) in registers.
+  const TileAddr ta = tile_addr<NT>(tid);
+  int tdst[2][8];
+  transposed_dst<NT>(tid, tdst);
+  int af_off[4][2], bf_off[4][2];
+#pragma unroll
+  for (int t16 = 0; t16 < 4; ++t16)
+#pragma unroll
+    for (int ds = 0; ds < 2; ++ds) {
+      af_off[t16][ds] = swz(t16 * 16 + c, (ds * 32 + 8 * g) * 2);
+      bf_off[t16][ds] = af_off[t16][ds];
+    }
+
+  // This wave's K (prescaled by scale*log2e) and V fragments in registers.
+  const float ks_scale = scale * LOG2E;
+  const int key_lo = jb * BK + w * 16;   // first key of this wave
+  const int key_me = key_lo + c;         // this lane's key
   bfrag k_frag[2], v_frag[2];
 #pragma unroll
   for (int ds = 0; ds < 2; ++ds) {
-    k_frag[ds] = *reinterpret_cast<const bfrag*>(
-        kp + (jb * BLK + w * 16 + c) * D + ds * 32 + 8 * g);
-    v_frag[ds] = *reinterpret_cast<const bfrag*>(
-        vp + (jb * BLK + w * 16 + c) * D + ds * 32 + 8 * g);
+    k_frag[ds] = load_frag_scaled(kp, jb * BK + w * 16 + c, ds * 32 + 8 * g,
+                                  ks_scale);
+    v_frag[ds] = load_frag(vp, jb * BK + w * 16 + c, ds * 32 + 8 * g);
   }
 
   f32x4 dk_acc[4] = {};
   f32x4 dv_acc[4] = {};
 
-  for (int i = jb; i < T / BLK; ++i) {
+  for (int i = jb * BK / KVB; i < T / KVB; ++i) {
     __syncthreads();
-    stage_rowmajor(qp + i * BLK * D, lds_q);
-    stage_rowmajor(dop + i * BLK * D, lds_do);
-    stage_transposed(qp + i * BLK * D, lds_qt);
-    stage_transposed(dop + i * BLK * D, lds_dot);
-    if (threadIdx.x < BLK) {
-      lds_lse[threadIdx.x] = lse[bh * T + i * BLK + threadIdx.x];
-      lds_dlt[threadIdx.x] = delta[bh * T + i * BLK + threadIdx.x];
+    stage_rowmajor<NT>(qp + i * KVB * D, lds_q, ta);
+    stage_rowmajor<NT>(dop + i * KVB * D, lds_do, ta);
+    stage_transposed<NT>(qp + i * KVB * D, lds_qt, ta, tdst);
+    stage_transposed<NT>(dop + i * KVB * D, lds_dot, ta, tdst);
+    if (tid < KVB) {
+      lds_lse[tid] = lse[bh * T + i * KVB + tid] * LOG2E;
+      lds_dlt[tid] = delta[bh * T + i * KVB + tid];
     }
     __syncthreads();
 
-    // S and dP tiles: C[qrow = 16*qt + 4g + r][key = w*16 + c]
-    unsigned pwp[4][2];   // P packed
-    unsigned pws[4][2];   // dS packed
+    const int q0 = i * KVB;
+    if (q0 + KVB - 1 < key_lo) continue;   // all rows above this wave's keys
+
+    const bool diag = q0 < key_lo + 16;    // some row may precede a key
+    unsigned pwp[4][2];
+    unsigned pws[4][2];
 #pragma unroll
     for (int qt = 0; qt < 4; ++qt) {
       f32x4 s_acc = {};
       f32x4 dp_acc = {};
-#pragma unroll
-      for (int ds = 0; ds < 2; ++ds) {
-        bfrag aq = frag_read(lds_q, qt * 16, ds * 32, g, c);
-        bfrag ado = frag_read(lds_do, qt * 16, ds * 32, g, c);
-        s_acc = MFMA_BF16_16x16x32(aq, k_frag[ds], s_acc);
-        dp_acc = MFMA_BF16_16x16x32(ado, v_frag[ds], dp_acc);
-      }
+      s_acc = MFMA_BF16_16x16x32(lds_read16(lds_q, af_off[qt][0]), k_frag[0], s_acc);
+      s_acc = MFMA_BF16_16x16x32(lds_read16(lds_q, af_off[qt][1]), k_frag[1], s_acc);
+      dp_acc = MFMA_BF16_16x16x32(lds_read16(lds_do, af_off[qt][0]), v_frag[0], dp_acc);
+      dp_acc = MFMA_BF16_16x16x32(lds_read16(lds_do, af_off[qt][1]), v_frag[1], dp_acc);
       float p[4], dsv[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int qrow = i * BLK + qt * 16 + 4 * g + r;
-        const int key = jb * BLK + w * 16 + c;
-        const float l = lds_lse[qt * 16 + 4 * g + r];
+        const int qrow = q0 + qt * 16 + 4 * g + r;
+        const float l2 = lds_lse[qt * 16 + 4 * g + r];
         const float dlt = lds_dlt[qt * 16 + 4 * g + r];
-        float pp = (key <= qrow) ? __expf(s_acc[r] * scale - l) : 0.f;
+        float pp = exp2f(s_acc[r] - l2);
+        if (diag) pp = (key_me <= qrow) ? pp : 0.f;
         p[r] = pp;
-        dsv[r] = scale * pp * (dp_acc[r] - dlt);
+        dsv[r] = pp * (dp_acc[r] - dlt);   // scale folded into epilogue
       }
       pwp[qt][0] = pack2(p[0], p[1]);
       pwp[qt][1] = pack2(p[2], p[3]);
@@ -355,44 +432,37 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
       pws[qt][1] = pack2(dsv[2], dsv[3]);
     }
 
-    // dV[key][d] += P^T dO ; dK[key][d] += dS^T Q
     const CShuffled pshp = cshuffle(pwp, g, c);
     const CShuffled pshs = cshuffle(pws, g, c);
-    {
-      bfrag ap0 = frag_from_shuffled<0>(pshp, g);
-      bfrag ap1 = frag_from_shuffled<1>(pshp, g);
-      bfrag as0 = frag_from_shuffled<0>(pshs, g);
-      bfrag as1 = frag_from_shuffled<1>(pshs, g);
+    const bfrag ap0 = frag_from_shuffled<0>(pshp, g);
+    const bfrag ap1 = frag_from_shuffled<1>(pshp, g);
+    const bfrag as0 = frag_from_shuffled<0>(pshs, g);
+    const bfrag as1 = frag_from_shuffled<1>(pshs, g);
 #pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-        bfrag bdo0 = frag_read(lds_dot, dt * 16, 0, g, c);
-        bfrag bdo1 = frag_read(lds_dot, dt * 16, 32, g, c);
-        bfrag bq0 = frag_read(lds_qt, dt * 16, 0, g, c);
-        bfrag bq1 = frag_read(lds_qt, dt * 16, 32, g, c);
-        dv_acc[dt] = MFMA_BF16_16x16x32(ap0, bdo0, dv_acc[dt]);
-        dv_acc[dt] = MFMA_BF16_16x16x32(ap1, bdo1, dv_acc[dt]);
-        dk_acc[dt] = MFMA_BF16_16x16x32(as0, bq0, dk_acc[dt]);
-        dk_acc[dt] = MFMA_BF16_16x16x32(as1, bq1, dk_acc[dt]);
-      }
+    for (int dt = 0; dt < 4; ++dt) {
+      dv_acc[dt] = MFMA_BF16_16x16x32(ap0, lds_read16(lds_dot, bf_off[dt][0]), dv_acc[dt]);
+      dv_acc[dt] = MFMA_BF16_16x16x32(ap1, lds_read16(lds_dot, bf_off[dt][1]), dv_acc[dt]);
+      dk_acc[dt] = MFMA_BF16_16x16x32(as0, lds_read16(lds_qt, bf_off[dt][0]), dk_acc[dt]);
+      dk_acc[dt] = MFMA_BF16_16x16x32(as1, lds_read16(lds_qt, bf_off[dt][1]), dk_acc[dt]);
     }
   }
 
-  // write: lane (g,c) holds rows key = w*16 + 4g + r, col = dt*16 + c
-  bf16* dkp = dk + (bh * T + jb * BLK + w * 16) * D;
-  bf16* dvp = dv + (bh * T + jb * BLK + w * 16) * D;
+  bf16* dkp = dk + (bh * T + jb * BK + w * 16) * D;
+  bf16* dvp = dv + (bh * T + jb * BK + w * 16) * D;
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt)
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      dkp[(4 * g + r) * D + dt * 16 + c] = f2bf(dk_acc[dt][r]);
+      dkp[(4 * g + r) * D + dt * 16 + c] = f2bf(dk_acc[dt][r] * scale);
       dvp[(4 * g + r) * D + dt * 16 + c] = f2bf(dv_acc[dt][r]);
     }
 }
 
 // ---------------------------------------------------------------------------
-// Backward dQ: one workgroup per 64-row Q block; wave w owns rows w*16..+15.
+// Backward dQ: one workgroup per (NW*16)-row Q block; wave w owns 16 rows.
 // ---------------------------------------------------------------------------
-__launch_bounds__(256, 2)
+template <int NW>
+__launch_bounds__(NW * WAVE, 2)
 __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
                                    const bf16* __restrict__ k,
                                    const bf16* __restrict__ v,
@@ -400,90 +470,99 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
                                    const float* __restrict__ lse,
                                    const float* __restrict__ delta,
                                    bf16* __restrict__ dq, int T, float scale) {
-  __shared__ __attribute__((aligned(16))) char smem[3 * BLK * D * 2];
+  constexpr int BM = NW * 16;
+  constexpr int NT = NW * WAVE;
+  __shared__ __attribute__((aligned(16))) char smem[3 * KVB * D * 2];
   char* lds_k = smem;                     // K row-major (A of S^T)
-  char* lds_kt = smem + BLK * D * 2;      // K^T (B of dQ)
-  char* lds_v = smem + 2 * BLK * D * 2;   // V row-major (A of dP^T)
+  char* lds_kt = smem + KVB * D * 2;      // K^T (B of dQ)
+  char* lds_v = smem + 2 * KVB * D * 2;   // V row-major (A of dP^T)
 
   const int qb = blockIdx.x;
   const long long bh = blockIdx.y;
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int w = threadIdx.x / WAVE;
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid / WAVE;
   const int g = lane >> 4;
   const int c = lane & 15;
 
-  const bf16* qp = q + (bh * T + qb * BLK) * D;
+  const bf16* qp = q + (bh * T + qb * BM) * D;
   const bf16* kp = k + bh * T * D;
   const bf16* vp = v + bh * T * D;
-  const bf16* dop = dout + (bh * T + qb * BLK) * D;
+  const bf16* dop = dout + (bh * T + qb * BM) * D;
 
+  const TileAddr ta = tile_addr<NT>(tid);
+  int tdst[2][8];
+  transposed_dst<NT>(tid, tdst);
+  int f_off[4][2];
+#pragma unroll
+  for (int t16 = 0; t16 < 4; ++t16)
+#pragma unroll
+    for (int ds = 0; ds < 2; ++ds)
+      f_off[t16][ds] = swz(t16 * 16 + c, (ds * 32 + 8 * g) * 2);
+
+  const float qs = scale * LOG2E;
   bfrag q_frag[2], do_frag[2];
 #pragma unroll
   for (int ds = 0; ds < 2; ++ds) {
-    q_frag[ds] = *reinterpret_cast<const bfrag*>(
-        qp + (w * 16 + c) * D + ds * 32 + 8 * g);
-    do_frag[ds] = *reinterpret_cast<const bfrag*>(
-        dop + (w * 16 + c) * D + ds * 32 + 8 * g);
+    q_frag[ds] = load_frag_scaled(qp, w * 16 + c, ds * 32 + 8 * g, qs);
+    do_frag[ds] = load_frag(dop, w * 16 + c, ds * 32 + 8 * g);
   }
-  const int qrow_mine = qb * BLK + w * 16 + c;
-  const float lse_mine = lse[bh * T + qrow_mine];
-  const float dlt_mine = delta[bh * T + qrow_mine];
+  const int row_lo = qb * BM + w * 16;
+  const int row_me = row_lo + c;
+  const float lse2_me = lse[bh * T + row_me] * LOG2E;
+  const float dlt_me = delta[bh * T + row_me];
 
   f32x4 dq_acc[4] = {};
 
-  for (int j = 0; j <= qb; ++j) {
+  const int n_kv = (qb + 1) * BM / KVB;
+  for (int j = 0; j < n_kv; ++j) {
     __syncthreads();
-    stage_rowmajor(kp + j * BLK * D, lds_k);
-    stage_transposed(kp + j * BLK * D, lds_kt);
-    stage_rowmajor(vp + j * BLK * D, lds_v);
+    stage_rowmajor<NT>(kp + j * KVB * D, lds_k, ta);
+    stage_transposed<NT>(kp + j * KVB * D, lds_kt, ta, tdst);
+    stage_rowmajor<NT>(vp + j * KVB * D, lds_v, ta);
     __syncthreads();
 
-    // S^T and dP^T tiles: C[key = 16*sub + 4g + r][qrow = c]
-    unsigned pws[4][2];  // dS^T packed (k-index = key)
+    const int key0 = j * KVB;
+    if (key0 > row_lo + 15) continue;
+
+    const bool diag = key0 + KVB - 1 > row_lo;
+    unsigned pws[4][2];
 #pragma unroll
     for (int sub = 0; sub < 4; ++sub) {
       f32x4 s_acc = {};
       f32x4 dp_acc = {};
-#pragma unroll
-      for (int ds = 0; ds < 2; ++ds) {
-        bfrag ak = frag_read(lds_k, sub * 16, ds * 32, g, c);
-        bfrag av = frag_read(lds_v, sub * 16, ds * 32, g, c);
-        s_acc = MFMA_BF16_16x16x32(ak, q_frag[ds], s_acc);
-        dp_acc = MFMA_BF16_16x16x32(av, do_frag[ds], dp_acc);
-      }
+      s_acc = MFMA_BF16_16x16x32(lds_read16(lds_k, f_off[sub][0]), q_frag[0], s_acc);
+      s_acc = MFMA_BF16_16x16x32(lds_read16(lds_k, f_off[sub][1]), q_frag[1], s_acc);
+      dp_acc = MFMA_BF16_16x16x32(lds_read16(lds_v, f_off[sub][0]), do_frag[0], dp_acc);
+      dp_acc = MFMA_BF16_16x16x32(lds_read16(lds_v, f_off[sub][1]), do_frag[1], dp_acc);
       float dsv[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int key = j * BLK + sub * 16 + 4 * g + r;
-        float pp = (key <= qrow_mine)
-                       ? __expf(s_acc[r] * scale - lse_mine) : 0.f;
-        dsv[r] = scale * pp * (dp_acc[r] - dlt_mine);
+        const int key = key0 + sub * 16 + 4 * g + r;
+        float pp = exp2f(s_acc[r] - lse2_me);
+        if (diag) pp = (key <= row_me) ? pp : 0.f;
+        dsv[r] = pp * (dp_acc[r] - dlt_me);  // scale folded into epilogue
       }
       pws[sub][0] = pack2(dsv[0], dsv[1]);
       pws[sub][1] = pack2(dsv[2], dsv[3]);
     }
 
-    // dQ[qrow][d] += dS K
     const CShuffled pshs = cshuffle(pws, g, c);
-    {
-      bfrag as0 = frag_from_shuffled<0>(pshs, g);
-      bfrag as1 = frag_from_shuffled<1>(pshs, g);
+    const bfrag as0 = frag_from_shuffled<0>(pshs, g);
+    const bfrag as1 = frag_from_shuffled<1>(pshs, g);
 #pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-        bfrag bk0 = frag_read(lds_kt, dt * 16, 0, g, c);
-        bfrag bk1 = frag_read(lds_kt, dt * 16, 32, g, c);
-        dq_acc[dt] = MFMA_BF16_16x16x32(as0, bk0, dq_acc[dt]);
-        dq_acc[dt] = MFMA_BF16_16x16x32(as1, bk1, dq_acc[dt]);
-      }
+    for (int dt = 0; dt < 4; ++dt) {
+      dq_acc[dt] = MFMA_BF16_16x16x32(as0, lds_read16(lds_kt, f_off[dt][0]), dq_acc[dt]);
+      dq_acc[dt] = MFMA_BF16_16x16x32(as1, lds_read16(lds_kt, f_off[dt][1]), dq_acc[dt]);
     }
   }
 
-  bf16* dqp = dq + (bh * T + qb * BLK + w * 16) * D;
+  bf16* dqp = dq + (bh * T + qb * BM + w * 16) * D;
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt)
 #pragma unroll
     for (int r = 0; r < 4; ++r)
-      dqp[(4 * g + r) * D + dt * 16 + c] = f2bf(dq_acc[dt][r]);
+      dqp[(4 * g + r) * D + dt * 16 + c] = f2bf(dq_acc[dt][r] * scale);
 }
 
 }  // namespace
@@ -493,11 +572,18 @@ extern "C" {
 hipError_t tdsa_attn_fwd(const void* q, const void* k, const void* v, void* o,
                          float* lse, long long BH, int T, float scale,
                          hipStream_t stream) {
-  if (T % BLK) return hipErrorInvalidValue;
-  dim3 grid(T / BLK, BH);
-  hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(256), 0, stream,
-                     (const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o,
-                     lse, T, scale);
+  if (T % KVB) return hipErrorInvalidValue;
+  if (T % 128 == 0) {
+    dim3 grid(T / 128, BH);
+    hipLaunchKernelGGL(attn_fwd_kernel<8>, grid, dim3(512), 0, stream,
+                       (const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o,
+                       lse, T, scale);
+  } else {
+    dim3 grid(T / 64, BH);
+    hipLaunchKernelGGL(attn_fwd_kernel<4>, grid, dim3(256), 0, stream,
+                       (const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o,
+                       lse, T, scale);
+  }
   return hipGetLastError();
 }
 
@@ -505,7 +591,7 @@ hipError_t tdsa_attn_bwd(const void* q, const void* k, const void* v,
                          const void* o, const float* lse, const void* dout,
                          void* dq, void* dk, void* dv, float* delta,
                          long long BH, int T, float scale, hipStream_t stream) {
-  if (T % BLK) return hipErrorInvalidValue;
+  if (T % KVB) return hipErrorInvalidValue;
   const long long R = BH * T;
   {
     const int rows_per_block = 256 / WAVE;
@@ -513,14 +599,25 @@ hipError_t tdsa_attn_bwd(const void* q, const void* k, const void* v,
     hipLaunchKernelGGL(attn_delta_kernel, dim3(grid), dim3(256), 0, stream,
                        (const bf16*)dout, (const bf16*)o, delta, R);
   }
-  dim3 grid(T / BLK, BH);
-  hipLaunchKernelGGL(attn_bwd_dkv_kernel, grid, dim3(256), 0, stream,
-                     (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                     (const bf16*)dout, lse, delta, (bf16*)dk, (bf16*)dv, T,
-                     scale);
-  hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, dim3(256), 0, stream,
-                     (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                     (const bf16*)dout, lse, delta, (bf16*)dq, T, scale);
+  if (T % 128 == 0) {
+    dim3 grid(T / 128, BH);
+    hipLaunchKernelGGL(attn_bwd_dkv_kernel<8>, grid, dim3(512), 0, stream,
+                       (const bf16*)q, (const bf16*)k, (const bf16*)v,
+                       (const bf16*)dout, lse, delta, (bf16*)dk, (bf16*)dv, T,
+                       scale);
+    hipLaunchKernelGGL(attn_bwd_dq_kernel<8>, grid, dim3(512), 0, stream,
+                       (const bf16*)q, (const bf16*)k, (const bf16*)v,
+                       (const bf16*)dout, lse, delta, (bf16*)dq, T, scale);
+  } else {
+    dim3 grid(T / 64, BH);
+    hipLaunchKernelGGL(attn_bwd_dkv_kernel<4>, grid, dim3(256), 0, stream,
+                       (const bf16*)q, (const bf16*)k, (const bf16*)v,
+                       (const bf16*)dout, lse, delta, (bf16*)dk, (bf16*)dv, T,
+                       scale);
+    hipLaunchKernelGGL(attn_bwd_dq_kernel<4>, grid, dim3(256), 0, stream,
+                       (const bf16*)q, (const bf16*)k, (const bf16*)v,
+                       (const bf16*)dout, lse, delta, (bf16*)dq, T, scale);
+  }
   return hipGetLastError();
 }
 
