@@ -7,6 +7,8 @@
 
 #include <hip/hip_runtime_api.h>
 
+#include <array>
+
 // ---- extern "C" launcher prototypes (csrc/kernels/*.hip) ------------------
 extern "C" {
 hipError_t tdsa_ln_fwd(const void*, const void*, const void*, void*, float*,
@@ -37,10 +39,12 @@ hipError_t tdsa_sgd_step(void*, const void*, float*, float*, int, int, float,
                          float, float, float, int, int, int, long long, int,
                          int, hipStream_t);
 hipError_t tdsa_attn_fwd(const void*, const void*, const void*, void*, float*,
-                         long long, int, float, hipStream_t);
+                         long long, long long, int, float, const long long*,
+                         const long long*, hipStream_t);
 hipError_t tdsa_attn_bwd(const void*, const void*, const void*, const void*,
                          const float*, const void*, void*, void*, void*, float*,
-                         long long, int, float, hipStream_t);
+                         long long, long long, int, float, const long long*,
+                         const long long*, const long long*, hipStream_t);
 hipError_t tdsa_dbg_mfma(const void*, const void*, float*, int, hipStream_t);
 hipError_t tdsa_dbg_stage(const void*, void*, int, hipStream_t);
 }
@@ -251,41 +255,69 @@ void sgd_step(at::Tensor param, at::Tensor grad, at::Tensor buf,
 }
 
 // ---- attention ------------------------------------------------------------
+static void check_attn_tensor(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.dim() == 4 && t.size(3) == 64, name,
+              ": attention kernel requires (B,H,T,64)");
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.stride(3) == 1, name, " last dim must be contiguous");
+}
+
+static std::array<long long, 3> strides3(const at::Tensor& t) {
+  return {(long long)t.stride(0), (long long)t.stride(1),
+          (long long)t.stride(2)};
+}
+
 std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
-                                      double scale) {
-  CHECK_IN(q); CHECK_IN(k); CHECK_IN(v);
-  TORCH_CHECK(q.dim() == 4 && q.size(3) == 64,
-              "attention kernel requires (B,H,T,64)");
-  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attention kernel is bf16");
-  TORCH_CHECK(q.size(2) % 64 == 0, "attention kernel requires T % 64 == 0");
-  const long long BH = (long long)q.size(0) * q.size(1);
+                                      double scale,
+                                      c10::optional<at::Tensor> out) {
+  check_attn_tensor(q, "q");
+  check_attn_tensor(k, "k");
+  check_attn_tensor(v, "v");
+  TORCH_CHECK(q.strides() == k.strides() && q.strides() == v.strides(),
+              "q/k/v must share strides");
+  TORCH_CHECK(q.size(2) % 64 == 0, "attention kernel requires T %% 64 == 0");
+  const long long B = q.size(0), H = q.size(1);
   const int T = q.size(2);
-  auto o = at::empty_like(q);
-  auto lse = at::empty({q.size(0), q.size(1), q.size(2)},
-                       q.options().dtype(at::kFloat));
+  at::Tensor o = out.has_value() ? *out : at::empty_like(q);
+  check_attn_tensor(o, "o");
+  auto lse = at::empty({B, H, (long long)T}, q.options().dtype(at::kFloat));
+  auto sq = strides3(q);
+  auto so = strides3(o);
   check_hip(tdsa_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
-                          o.data_ptr(), lse.data_ptr<float>(), BH, T,
-                          (float)scale, cur_stream()),
+                          o.data_ptr(), lse.data_ptr<float>(), B, H, T,
+                          (float)scale, sq.data(), so.data(), cur_stream()),
             "attention_fwd");
   return {o, lse};
 }
 
 std::vector<at::Tensor> attention_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                       at::Tensor o, at::Tensor lse,
-                                      at::Tensor dout, double scale) {
-  CHECK_IN(q); CHECK_IN(k); CHECK_IN(v); CHECK_IN(o); CHECK_IN(dout);
+                                      at::Tensor dout, double scale,
+                                      c10::optional<at::Tensor> dq_out,
+                                      c10::optional<at::Tensor> dk_out,
+                                      c10::optional<at::Tensor> dv_out) {
+  check_attn_tensor(q, "q");
+  check_attn_tensor(o, "o");
+  check_attn_tensor(dout, "dout");
+  TORCH_CHECK(o.strides() == dout.strides(), "o/dout must share strides");
   auto lsec = lse.contiguous();
-  const long long BH = (long long)q.size(0) * q.size(1);
+  const long long B = q.size(0), H = q.size(1);
   const int T = q.size(2);
-  auto dq = at::empty_like(q);
-  auto dk = at::empty_like(k);
-  auto dv = at::empty_like(v);
-  auto delta = at::empty({BH * T}, q.options().dtype(at::kFloat));
+  at::Tensor dq = dq_out.has_value() ? *dq_out : at::empty_like(q);
+  at::Tensor dk = dk_out.has_value() ? *dk_out : at::empty_like(k);
+  at::Tensor dv = dv_out.has_value() ? *dv_out : at::empty_like(v);
+  TORCH_CHECK(dq.strides() == dk.strides() && dq.strides() == dv.strides(),
+              "dq/dk/dv must share strides");
+  auto delta = at::empty({B * H * T}, q.options().dtype(at::kFloat));
+  auto sq = strides3(q);
+  auto so = strides3(o);
+  auto sd = strides3(dq);
   check_hip(tdsa_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                           o.data_ptr(), lsec.data_ptr<float>(), dout.data_ptr(),
                           dq.data_ptr(), dk.data_ptr(), dv.data_ptr(),
-                          delta.data_ptr<float>(), BH, T, (float)scale,
-                          cur_stream()),
+                          delta.data_ptr<float>(), B, H, T, (float)scale,
+                          sq.data(), so.data(), sd.data(), cur_stream()),
             "attention_bwd");
   return {dq, dk, dv};
 }
@@ -326,6 +358,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("cross_entropy_bwd", &cross_entropy_bwd);
   mod.def("adamw_step", &adamw_step);
   mod.def("sgd_step", &sgd_step);
-  mod.def("attention_fwd", &attention_fwd);
-  mod.def("attention_bwd", &attention_bwd);
+  mod.def("attention_fwd", &attention_fwd, py::arg("q"), py::arg("k"),
+          py::arg("v"), py::arg("scale"), py::arg("out") = py::none());
+  mod.def("attention_bwd", &attention_bwd, py::arg("q"), py::arg("k"),
+          py::arg("v"), py::arg("o"), py::arg("lse"), py::arg("dout"),
+          py::arg("scale"), py::arg("dq") = py::none(),
+          py::arg("dk") = py::none(), py::arg("dv") = py::none());
 }

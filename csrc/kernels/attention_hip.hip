@@ -61,6 +61,15 @@ DEV_INLINE bfrag lds_read16(const char* lds, int byte_off) {
   return *reinterpret_cast<const bfrag*>(lds + byte_off);
 }
 
+// Per-tensor global strides in ELEMENTS (last dim must be contiguous).
+// Lets the kernels consume the packed qkv projection / (B,T,H,D) activation
+// layouts directly — no transpose-copies on the hot path.
+struct GStride {
+  long long b;  // batch stride
+  long long h;  // head stride
+  int t;        // row (token) stride
+};
+
 // Per-lane precomputed addressing for one [64][64] staging + fragment-read
 // pattern. Staging is 512 chunks of 8 bf16 over NT threads.
 struct TileAddr {
@@ -69,12 +78,12 @@ struct TileAddr {
 };
 
 template <int NT>
-DEV_INLINE TileAddr tile_addr(int tid) {
+DEV_INLINE TileAddr tile_addr(int tid, int t_stride) {
   TileAddr a;
 #pragma unroll
   for (int rep = 0; rep < 512 / (NT / 8) / 8; ++rep) {
     int chunk = tid + rep * NT;
-    a.stage_src[rep] = chunk * 8;
+    a.stage_src[rep] = (chunk >> 3) * t_stride + (chunk & 7) * 8;
     a.stage_dst[rep] = swz(chunk >> 3, (chunk & 7) * 16);
   }
   return a;
@@ -144,17 +153,18 @@ DEV_INLINE bfrag frag_from_shuffled(const CShuffled& s, int g) {
 }
 
 // load a global B/A-style fragment (row r0+c, 8 elems at col c0+8g) scaled
-DEV_INLINE bfrag load_frag_scaled(const bf16* p, int row, int col, float s) {
-  short8v v = load8(p + row * D + col);
+DEV_INLINE bfrag load_frag_scaled(const bf16* p, int row, int col, int st,
+                                  float s) {
+  short8v v = load8(p + (long long)row * st + col);
   union { bfrag f; short w[8]; } r;
 #pragma unroll
   for (int e = 0; e < 8; ++e) r.w[e] = bf_pack(bf_elem(v, e) * s);
   return r.f;
 }
 
-DEV_INLINE bfrag load_frag(const bf16* p, int row, int col) {
+DEV_INLINE bfrag load_frag(const bf16* p, int row, int col, int st) {
   union { bfrag f; short8v v; } r;
-  r.v = load8(p + row * D + col);
+  r.v = load8(p + (long long)row * st + col);
   return r.f;
 }
 
@@ -167,7 +177,8 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
                                 const bf16* __restrict__ k,
                                 const bf16* __restrict__ v,
                                 bf16* __restrict__ o, float* __restrict__ lse,
-                                int T, float scale) {
+                                int T, int H, float scale, GStride sq,
+                                GStride so) {
   constexpr int BM = NW * 16;  // q rows per workgroup
   constexpr int NT = NW * WAVE;
   __shared__ __attribute__((aligned(16))) char smem[2 * KVB * D * 2];
@@ -182,12 +193,13 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
   const int g = lane >> 4;
   const int c = lane & 15;
 
-  const bf16* qp = q + (bh * T + qb * BM) * D;
-  const bf16* kp = k + bh * T * D;
-  const bf16* vp = v + bh * T * D;
+  const long long boff = (bh / H) * sq.b + (bh % H) * sq.h;
+  const bf16* qp = q + boff + (long long)(qb * BM) * sq.t;
+  const bf16* kp = k + boff;
+  const bf16* vp = v + boff;
 
   // addressing, hoisted out of the K/V loop
-  const TileAddr ta = tile_addr<NT>(tid);
+  const TileAddr ta = tile_addr<NT>(tid, sq.t);
   int tdst[2][8];
   transposed_dst<NT>(tid, tdst);
   int kf_off[4][2], vf_off[4][2];
@@ -204,7 +216,7 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
   bfrag q_frag[2];
 #pragma unroll
   for (int ds = 0; ds < 2; ++ds)
-    q_frag[ds] = load_frag_scaled(qp, w * 16 + c, ds * 32 + 8 * g, qs);
+    q_frag[ds] = load_frag_scaled(qp, w * 16 + c, ds * 32 + 8 * g, sq.t, qs);
 
   f32x4 o_acc[4] = {};
   float m_run = -INFINITY;
@@ -215,8 +227,8 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
   const int n_kv = (qb + 1) * BM / KVB;
   for (int j = 0; j < n_kv; ++j) {
     __syncthreads();
-    stage_rowmajor<NT>(kp + j * KVB * D, lds_k, ta);
-    stage_transposed<NT>(vp + j * KVB * D, lds_vt, ta, tdst);
+    stage_rowmajor<NT>(kp + (long long)(j * KVB) * sq.t, lds_k, ta);
+    stage_transposed<NT>(vp + (long long)(j * KVB) * sq.t, lds_vt, ta, tdst);
     __syncthreads();
 
     const int key0 = j * KVB;
@@ -301,12 +313,13 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
 #pragma unroll
   for (int r = 0; r < 4; ++r)
     linv_row[r] = 1.0f / __shfl(l_run, 4 * g + r, WAVE);
-  bf16* op = o + (bh * T + qb * BM + w * 16) * D;
+  bf16* op = o + (bh / H) * so.b + (bh % H) * so.h
+             + (long long)(qb * BM + w * 16) * so.t;
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt)
 #pragma unroll
     for (int r = 0; r < 4; ++r)
-      op[(4 * g + r) * D + dt * 16 + c] = f2bf(o_acc[dt][r] * linv_row[r]);
+      op[(4 * g + r) * so.t + dt * 16 + c] = f2bf(o_acc[dt][r] * linv_row[r]);
   if (lane < 16) {
     lse[bh * T + qb * BM + w * 16 + c] = (m_run + log2f(l_run)) * LN2;
   }
@@ -317,12 +330,16 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
 // ---------------------------------------------------------------------------
 __global__ void attn_delta_kernel(const bf16* __restrict__ dout,
                                   const bf16* __restrict__ o,
-                                  float* __restrict__ delta, long long R) {
+                                  float* __restrict__ delta, long long R,
+                                  int T, int H, GStride so) {
   const long long row = (long long)blockIdx.x * (blockDim.x / WAVE)
                         + threadIdx.x / WAVE;
   if (row >= R) return;
   const int lane = threadIdx.x & (WAVE - 1);
-  float acc = bf2f(dout[row * D + lane]) * bf2f(o[row * D + lane]);
+  const long long bh = row / T;
+  const long long off = (bh / H) * so.b + (bh % H) * so.h
+                        + (long long)(row % T) * so.t + lane;
+  float acc = bf2f(dout[off]) * bf2f(o[off]);
   acc = wave_sum(acc);
   if (lane == 0) delta[row] = acc;
 }
@@ -339,7 +356,8 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
                                     const float* __restrict__ lse,
                                     const float* __restrict__ delta,
                                     bf16* __restrict__ dk, bf16* __restrict__ dv,
-                                    int T, float scale) {
+                                    int T, int H, float scale, GStride sq,
+                                    GStride so, GStride sd) {
   constexpr int BK = NW * 16;  // keys per workgroup
   constexpr int NT = NW * WAVE;
   __shared__ __attribute__((aligned(16))) char smem[4 * KVB * D * 2 + 2 * KVB * 4];
@@ -358,12 +376,15 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
   const int g = lane >> 4;
   const int c = lane & 15;
 
-  const bf16* qp = q + bh * T * D;
-  const bf16* kp = k + bh * T * D;
-  const bf16* vp = v + bh * T * D;
-  const bf16* dop = dout + bh * T * D;
+  const long long boff = (bh / H) * sq.b + (bh % H) * sq.h;
+  const long long ooff = (bh / H) * so.b + (bh % H) * so.h;
+  const bf16* qp = q + boff;
+  const bf16* kp = k + boff;
+  const bf16* vp = v + boff;
+  const bf16* dop = dout + ooff;
 
-  const TileAddr ta = tile_addr<NT>(tid);
+  const TileAddr ta = tile_addr<NT>(tid, sq.t);
+  const TileAddr tao = tile_addr<NT>(tid, so.t);
   int tdst[2][8];
   transposed_dst<NT>(tid, tdst);
   int af_off[4][2], bf_off[4][2];
@@ -383,8 +404,8 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
 #pragma unroll
   for (int ds = 0; ds < 2; ++ds) {
     k_frag[ds] = load_frag_scaled(kp, jb * BK + w * 16 + c, ds * 32 + 8 * g,
-                                  ks_scale);
-    v_frag[ds] = load_frag(vp, jb * BK + w * 16 + c, ds * 32 + 8 * g);
+                                  sq.t, ks_scale);
+    v_frag[ds] = load_frag(vp, jb * BK + w * 16 + c, ds * 32 + 8 * g, sq.t);
   }
 
   f32x4 dk_acc[4] = {};
@@ -392,10 +413,10 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
 
   for (int i = jb * BK / KVB; i < T / KVB; ++i) {
     __syncthreads();
-    stage_rowmajor<NT>(qp + i * KVB * D, lds_q, ta);
-    stage_rowmajor<NT>(dop + i * KVB * D, lds_do, ta);
-    stage_transposed<NT>(qp + i * KVB * D, lds_qt, ta, tdst);
-    stage_transposed<NT>(dop + i * KVB * D, lds_dot, ta, tdst);
+    stage_rowmajor<NT>(qp + (long long)(i * KVB) * sq.t, lds_q, ta);
+    stage_rowmajor<NT>(dop + (long long)(i * KVB) * so.t, lds_do, tao);
+    stage_transposed<NT>(qp + (long long)(i * KVB) * sq.t, lds_qt, ta, tdst);
+    stage_transposed<NT>(dop + (long long)(i * KVB) * so.t, lds_dot, tao, tdst);
     if (tid < KVB) {
       lds_lse[tid] = lse[bh * T + i * KVB + tid] * LOG2E;
       lds_dlt[tid] = delta[bh * T + i * KVB + tid];
@@ -448,14 +469,16 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
     }
   }
 
-  bf16* dkp = dk + (bh * T + jb * BK + w * 16) * D;
-  bf16* dvp = dv + (bh * T + jb * BK + w * 16) * D;
+  const long long doff = (bh / H) * sd.b + (bh % H) * sd.h
+                         + (long long)(jb * BK + w * 16) * sd.t;
+  bf16* dkp = dk + doff;
+  bf16* dvp = dv + doff;
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt)
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      dkp[(4 * g + r) * D + dt * 16 + c] = f2bf(dk_acc[dt][r] * scale);
-      dvp[(4 * g + r) * D + dt * 16 + c] = f2bf(dv_acc[dt][r]);
+      dkp[(4 * g + r) * sd.t + dt * 16 + c] = f2bf(dk_acc[dt][r] * scale);
+      dvp[(4 * g + r) * sd.t + dt * 16 + c] = f2bf(dv_acc[dt][r]);
     }
 }
 
@@ -470,7 +493,9 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
                                    const bf16* __restrict__ dout,
                                    const float* __restrict__ lse,
                                    const float* __restrict__ delta,
-                                   bf16* __restrict__ dq, int T, float scale) {
+                                   bf16* __restrict__ dq, int T, int H,
+                                   float scale, GStride sq, GStride so,
+                                   GStride sd) {
   constexpr int BM = NW * 16;
   constexpr int NT = NW * WAVE;
   __shared__ __attribute__((aligned(16))) char smem[3 * KVB * D * 2];
@@ -486,12 +511,14 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   const int g = lane >> 4;
   const int c = lane & 15;
 
-  const bf16* qp = q + (bh * T + qb * BM) * D;
-  const bf16* kp = k + bh * T * D;
-  const bf16* vp = v + bh * T * D;
-  const bf16* dop = dout + (bh * T + qb * BM) * D;
+  const long long boff = (bh / H) * sq.b + (bh % H) * sq.h;
+  const long long ooff = (bh / H) * so.b + (bh % H) * so.h;
+  const bf16* qp = q + boff + (long long)(qb * BM) * sq.t;
+  const bf16* kp = k + boff;
+  const bf16* vp = v + boff;
+  const bf16* dop = dout + ooff + (long long)(qb * BM) * so.t;
 
-  const TileAddr ta = tile_addr<NT>(tid);
+  const TileAddr ta = tile_addr<NT>(tid, sq.t);
   int tdst[2][8];
   transposed_dst<NT>(tid, tdst);
   int f_off[4][2];
@@ -505,8 +532,8 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   bfrag q_frag[2], do_frag[2];
 #pragma unroll
   for (int ds = 0; ds < 2; ++ds) {
-    q_frag[ds] = load_frag_scaled(qp, w * 16 + c, ds * 32 + 8 * g, qs);
-    do_frag[ds] = load_frag(dop, w * 16 + c, ds * 32 + 8 * g);
+    q_frag[ds] = load_frag_scaled(qp, w * 16 + c, ds * 32 + 8 * g, sq.t, qs);
+    do_frag[ds] = load_frag(dop, w * 16 + c, ds * 32 + 8 * g, so.t);
   }
   const int row_lo = qb * BM + w * 16;
   const int row_me = row_lo + c;
@@ -518,9 +545,9 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   const int n_kv = (qb + 1) * BM / KVB;
   for (int j = 0; j < n_kv; ++j) {
     __syncthreads();
-    stage_rowmajor<NT>(kp + j * KVB * D, lds_k, ta);
-    stage_transposed<NT>(kp + j * KVB * D, lds_kt, ta, tdst);
-    stage_rowmajor<NT>(vp + j * KVB * D, lds_v, ta);
+    stage_rowmajor<NT>(kp + (long long)(j * KVB) * sq.t, lds_k, ta);
+    stage_transposed<NT>(kp + (long long)(j * KVB) * sq.t, lds_kt, ta, tdst);
+    stage_rowmajor<NT>(vp + (long long)(j * KVB) * sq.t, lds_v, ta);
     __syncthreads();
 
     const int key0 = j * KVB;
@@ -558,32 +585,38 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
     }
   }
 
-  bf16* dqp = dq + (bh * T + qb * BM + w * 16) * D;
+  bf16* dqp = dq + (bh / H) * sd.b + (bh % H) * sd.h
+              + (long long)(qb * BM + w * 16) * sd.t;
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt)
 #pragma unroll
     for (int r = 0; r < 4; ++r)
-      dqp[(4 * g + r) * D + dt * 16 + c] = f2bf(dq_acc[dt][r] * scale);
+      dqp[(4 * g + r) * sd.t + dt * 16 + c] = f2bf(dq_acc[dt][r] * scale);
 }
 
 }  // namespace
 
 extern "C" {
 
+// strides arrays: {b, h, t} in elements, per tensor group:
+// sq = q/k/v, so = o (and dO in bwd), sd = dq/dk/dv.
 hipError_t tdsa_attn_fwd(const void* q, const void* k, const void* v, void* o,
-                         float* lse, long long BH, int T, float scale,
-                         hipStream_t stream) {
+                         float* lse, long long B, long long H, int T,
+                         float scale, const long long* sq_in,
+                         const long long* so_in, hipStream_t stream) {
   if (T % KVB) return hipErrorInvalidValue;
+  GStride sq{sq_in[0], sq_in[1], (int)sq_in[2]};
+  GStride so{so_in[0], so_in[1], (int)so_in[2]};
   if (T % 128 == 0) {
-    dim3 grid(T / 128, BH);
+    dim3 grid(T / 128, B * H);
     hipLaunchKernelGGL(attn_fwd_kernel<8>, grid, dim3(512), 0, stream,
                        (const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o,
-                       lse, T, scale);
+                       lse, T, (int)H, scale, sq, so);
   } else {
-    dim3 grid(T / 64, BH);
+    dim3 grid(T / 64, B * H);
     hipLaunchKernelGGL(attn_fwd_kernel<4>, grid, dim3(256), 0, stream,
                        (const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o,
-                       lse, T, scale);
+                       lse, T, (int)H, scale, sq, so);
   }
   return hipGetLastError();
 }
@@ -591,33 +624,42 @@ hipError_t tdsa_attn_fwd(const void* q, const void* k, const void* v, void* o,
 hipError_t tdsa_attn_bwd(const void* q, const void* k, const void* v,
                          const void* o, const float* lse, const void* dout,
                          void* dq, void* dk, void* dv, float* delta,
-                         long long BH, int T, float scale, hipStream_t stream) {
+                         long long B, long long H, int T, float scale,
+                         const long long* sq_in, const long long* so_in,
+                         const long long* sd_in, hipStream_t stream) {
   if (T % KVB) return hipErrorInvalidValue;
+  GStride sq{sq_in[0], sq_in[1], (int)sq_in[2]};
+  GStride so{so_in[0], so_in[1], (int)so_in[2]};
+  GStride sd{sd_in[0], sd_in[1], (int)sd_in[2]};
+  const long long BH = B * H;
   const long long R = BH * T;
   {
     const int rows_per_block = 256 / WAVE;
     const long long grid = (R + rows_per_block - 1) / rows_per_block;
     hipLaunchKernelGGL(attn_delta_kernel, dim3(grid), dim3(256), 0, stream,
-                       (const bf16*)dout, (const bf16*)o, delta, R);
+                       (const bf16*)dout, (const bf16*)o, delta, R, T, (int)H,
+                       so);
   }
   if (T % 128 == 0) {
     dim3 grid(T / 128, BH);
     hipLaunchKernelGGL(attn_bwd_dkv_kernel<8>, grid, dim3(512), 0, stream,
                        (const bf16*)q, (const bf16*)k, (const bf16*)v,
                        (const bf16*)dout, lse, delta, (bf16*)dk, (bf16*)dv, T,
-                       scale);
+                       (int)H, scale, sq, so, sd);
     hipLaunchKernelGGL(attn_bwd_dq_kernel<8>, grid, dim3(512), 0, stream,
                        (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                       (const bf16*)dout, lse, delta, (bf16*)dq, T, scale);
+                       (const bf16*)dout, lse, delta, (bf16*)dq, T, (int)H,
+                       scale, sq, so, sd);
   } else {
     dim3 grid(T / 64, BH);
     hipLaunchKernelGGL(attn_bwd_dkv_kernel<4>, grid, dim3(256), 0, stream,
                        (const bf16*)q, (const bf16*)k, (const bf16*)v,
                        (const bf16*)dout, lse, delta, (bf16*)dk, (bf16*)dv, T,
-                       scale);
+                       (int)H, scale, sq, so, sd);
     hipLaunchKernelGGL(attn_bwd_dq_kernel<4>, grid, dim3(256), 0, stream,
                        (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                       (const bf16*)dout, lse, delta, (bf16*)dq, T, scale);
+                       (const bf16*)dout, lse, delta, (bf16*)dq, T, (int)H,
+                       scale, sq, so, sd);
   }
   return hipGetLastError();
 }

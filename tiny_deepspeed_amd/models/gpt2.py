@@ -76,15 +76,19 @@ class CausalSelfAttention(nn.Module):
     def forward(self, x):
         B, T, E = x.shape
         qkv = self.c_attn(x)
+        if self.attention == "fused":
+            # packed-qkv path: the stride-aware kernel reads qkv directly,
+            # no transpose-copies (saves 8 (B,T,E)-sized copies per layer
+            # vs the reference's split+transpose+contiguous dance)
+            return self.c_proj(ops.fused_causal_attention(
+                qkv, self.n_head, dropout_p=self.dropout,
+                training=self.training))
         q, k, v = qkv.split(E, dim=2)
         # (B, T, E) -> (B, H, T, D)
         q = q.view(B, T, self.n_head, self.head_dim).transpose(1, 2).contiguous()
         k = k.view(B, T, self.n_head, self.head_dim).transpose(1, 2).contiguous()
         v = v.view(B, T, self.n_head, self.head_dim).transpose(1, 2).contiguous()
-        if self.attention == "fused":
-            y = ops.causal_attention(q, k, v, dropout_p=self.dropout,
-                                     training=self.training)
-        elif self.attention == "math":
+        if self.attention == "math":
             scale = 1.0 / math.sqrt(self.head_dim)
             mask = torch.ones(T, T, dtype=torch.bool, device=x.device).tril()
             s = torch.matmul(q, k.transpose(-2, -1)) * scale

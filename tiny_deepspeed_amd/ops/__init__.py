@@ -15,7 +15,7 @@ from .linear import (
 from .layernorm import layernorm_fwd, layernorm_dx, layernorm_dwdb
 from .embedding import embedding_forward, embedding_weight_grad
 from .gelu import gelu, gelu_fwd, gelu_bwd
-from .attention import causal_attention
+from .attention import causal_attention, fused_causal_attention
 from .cross_entropy import cross_entropy
 from .optim_ops import adamw_step, sgd_step
 from .autotuner import RuntimeAutoTuner
@@ -27,7 +27,7 @@ __all__ = [
     "layernorm_fwd", "layernorm_dx", "layernorm_dwdb",
     "embedding_forward", "embedding_weight_grad",
     "gelu", "gelu_fwd", "gelu_bwd",
-    "causal_attention",
+    "causal_attention", "fused_causal_attention",
     "cross_entropy",
     "adamw_step", "sgd_step",
     "RuntimeAutoTuner",
