@@ -1,43 +1,47 @@
 // Fused causal attention (flash-style) on CDNA4 MFMA matrix cores.
 //
 // Replaces the reference's materialized (B,H,T,T) score path
-// (/root/reference/example/model.py:29-51) with an MI355X-native design:
-//   forward : per Q block (8 waves x 16 rows), stream 64-key K/V tiles
-//             through XOR-swizzled LDS; S^T = mfma(K, Q) per 16x16 tile
-//             (v_mfma_f32_16x16x32_bf16, fp32 accum) so the softmax row is
-//             lane-local; online softmax in exp2 space (log2e folded into
-//             the Q prescale); P repacked to MFMA A-fragments with
-//             tile-uniform ds_bpermute shuffles; O = P V with V staged
-//             transposed. Saves per-row logsumexp (natural log).
+// (/root/reference/example/model.py:29-51) with an MI355X-native design
+// built around v_mfma_f32_32x32x16_bf16 (fp32 accumulate):
+//
+//   forward : per Q block (NW waves x 32 rows), stream 64-key K/V tiles
+//             through XOR-swizzled LDS. S^T = mfma(K, Q) puts the whole
+//             softmax row in one lane pair (one shfl_xor(32) per reduce);
+//             online softmax runs in exp2 space (log2e folded into the Q
+//             prescale); P is repacked into MFMA A-fragments with TWO
+//             v_permlane32_swap per k-slice (no LDS round trip); O = P V
+//             with V staged transposed. Saves per-row natural-log lse.
 //   backward: recompute-based two-kernel scheme (no atomics):
-//             dkv kernel owns a 128-key block and accumulates dK/dV over Q
-//             tiles; dq kernel owns a 128-row Q block. delta = rowsum(dO*O)
-//             by a small wave-reduction kernel. The 1/sqrt(D) factor on
-//             dS is folded into the dK/dQ epilogue.
+//             dkv kernel owns a (NW*32)-key block and accumulates dK/dV over
+//             64-row Q tiles; dq kernel owns a (NW*32)-row Q block.
+//             delta = rowsum(dO*O) by a small wave-reduction kernel.
+//             The 1/sqrt(D) on dS is folded into the dK/dQ epilogue.
 //
-// VALU-discipline (first profile showed 22:1 VALU:MFMA): every LDS offset
-// (fragment reads, staging stores) is computed once before the K/V loop;
-// causal masking runs only on diagonal tiles (wave-uniform branch); waves
-// whose rows lie entirely above/below a tile skip its compute.
+// All global accesses are stride-parameterized so the kernels consume the
+// packed qkv projection layout (B,T,3,H,D) and write O/dQKV into (B,T,H,D)
+// buffers directly — no transpose-copies on the hot path.
 //
-// Fragment layouts (gfx950, verified on hardware by scripts/debug_mfma.py
-// and against ck_tile/ops/gemm/warp/warp_gemm_attribute_mfma_impl.hpp):
-//   A[16x32] : lane l, elem e(0..7) -> A[l%16][(l/16)*8 + e]
-//   B[32x16] : lane l, elem e      -> B[(l/16)*8 + e][l%16]
-//   C[16x16] : lane l, reg  r(0..3)-> C[(l/16)*4 + r][l%16]
-// (The hardware pairs A/B by register slot, so A and B must agree on the
-// slot->k map; the C map is fixed.)
+// VALU discipline (first profile measured 22:1 VALU:MFMA on a naive
+// version): every LDS offset is precomputed before the K/V loop; causal
+// masking runs only on diagonal tiles (wave-uniform branch); waves whose
+// rows lie entirely outside a tile skip its compute.
 //
-// Contract: bf16 tensors (B,H,T,64) contiguous, T % 64 == 0. The Python op
-// (ops/attention.py) falls back to the composite path otherwise.
+// 32x32x16 fragment layouts (hardware-verified slot-pairing rule: A and B
+// must agree on the slot->k map, C is fixed; see scripts/debug_mfma.py):
+//   A[32x16]: lane l, elem e(0..7) -> A[l%32][(l/32)*8 + e]
+//   B[16x32]: lane l, elem e      -> B[(l/32)*8 + e][l%32]
+//   C[32x32]: lane l, reg r(0..15)-> C[(r&3) + 8*(r>>2) + 4*(l/32)][l%32]
+//
+// Contract: bf16, head_dim 64, T % 64 == 0 (the Python op falls back to a
+// composite rocBLAS path otherwise, ops/attention.py).
 #include "common.h"
 
-#define MFMA_BF16_16x16x32(a, b, c) \
-  __builtin_amdgcn_mfma_f32_16x16x32_bf16((a), (b), (c), 0, 0, 0)
+#define MFMA32(a, b, c) \
+  __builtin_amdgcn_mfma_f32_32x32x16_bf16((a), (b), (c), 0, 0, 0)
 
 namespace {
 
-constexpr int KVB = 64;   // kv keys per LDS tile
+constexpr int KVB = 64;   // keys / rows per LDS tile
 constexpr int D = 64;     // head dim (all GPT-2 sizes)
 constexpr float LOG2E = 1.4426950408889634f;
 constexpr float LN2 = 0.6931471805599453f;
@@ -60,27 +64,27 @@ DEV_INLINE bfrag lds_read16(const char* lds, int byte_off) {
   return *reinterpret_cast<const bfrag*>(lds + byte_off);
 }
 
-// Per-tensor global strides in ELEMENTS (last dim must be contiguous).
-// Lets the kernels consume the packed qkv projection / (B,T,H,D) activation
-// layouts directly — no transpose-copies on the hot path.
+// Row index of C register r for this half-wave (the 32x32 C layout).
+DEV_INLINE int crow(int r, int h32) { return (r & 3) + 8 * (r >> 2) + 4 * h32; }
+
+// Per-tensor global strides in ELEMENTS (last dim contiguous).
 struct GStride {
-  long long b;  // batch stride
-  long long h;  // head stride
-  int t;        // row (token) stride
+  long long b;
+  long long h;
+  int t;
 };
 
-// Per-lane precomputed addressing for one [64][64] staging + fragment-read
-// pattern. Staging is 512 chunks of 8 bf16 over NT threads.
+// Per-lane staging addresses for one [64][64] tile: 512 chunks of 8 bf16.
 struct TileAddr {
-  int stage_src[2];   // global element offsets of this thread's chunks
-  int stage_dst[2];   // swizzled LDS byte offsets (row-major image)
+  int stage_src[4];
+  int stage_dst[4];
 };
 
 template <int NT>
 DEV_INLINE TileAddr tile_addr(int tid, int t_stride) {
   TileAddr a;
 #pragma unroll
-  for (int rep = 0; rep < 512 / (NT / 8) / 8; ++rep) {
+  for (int rep = 0; rep < 512 / NT * 8 / 8; ++rep) {
     int chunk = tid + rep * NT;
     a.stage_src[rep] = (chunk >> 3) * t_stride + (chunk & 7) * 8;
     a.stage_dst[rep] = swz(chunk >> 3, (chunk & 7) * 16);
@@ -92,17 +96,17 @@ template <int NT>
 DEV_INLINE void stage_rowmajor(const bf16* __restrict__ g, char* lds,
                                const TileAddr& a) {
 #pragma unroll
-  for (int rep = 0; rep < 512 * 8 / NT / 8; ++rep)
+  for (int rep = 0; rep < 512 / NT; ++rep)
     *reinterpret_cast<short8v*>(lds + a.stage_dst[rep]) =
         load8(g + a.stage_src[rep]);
 }
 
-// Transposed staging: dest row = source col. Scalar u16 writes.
+// Transposed staging (dest row = source col), scalar u16 writes.
 template <int NT>
 DEV_INLINE void stage_transposed(const bf16* __restrict__ g, char* lds,
-                                 const TileAddr& a, const int (&tdst)[2][8]) {
+                                 const TileAddr& a, const int (&tdst)[4][8]) {
 #pragma unroll
-  for (int rep = 0; rep < 512 * 8 / NT / 8; ++rep) {
+  for (int rep = 0; rep < 512 / NT; ++rep) {
     short8v v = load8(g + a.stage_src[rep]);
 #pragma unroll
     for (int j = 0; j < 8; ++j)
@@ -111,9 +115,9 @@ DEV_INLINE void stage_transposed(const bf16* __restrict__ g, char* lds,
 }
 
 template <int NT>
-DEV_INLINE void transposed_dst(int tid, int (&tdst)[2][8]) {
+DEV_INLINE void transposed_dst(int tid, int (&tdst)[4][8]) {
 #pragma unroll
-  for (int rep = 0; rep < 512 * 8 / NT / 8; ++rep) {
+  for (int rep = 0; rep < 512 / NT; ++rep) {
     int chunk = tid + rep * NT;
     int row = chunk >> 3;
     int c0 = (chunk & 7) * 8;
@@ -122,36 +126,32 @@ DEV_INLINE void transposed_dst(int tid, int (&tdst)[2][8]) {
   }
 }
 
-// C-layout -> A-fragment redistribution (see header comment of cshuffle).
-struct CShuffled {
-  unsigned sh[4][2][2];  // [tile][word][half]
+// C-layout -> A-fragment repack. Values live per lane as 16 C registers per
+// 32-wide tile (packed to bf16 word pairs wA[r1]=(r0=0,1), wB[r1]=(r0=2,3)).
+// A-frag slice s (k = 16s + 8*h32 + e at this lane's own 32-axis index)
+// takes word pairs from C registers r1 = 2(s&1)+h of BOTH halves: one
+// permlane32_swap per word pair delivers own-half and partner-half at once.
+struct PackedC {
+  unsigned wA[2][4];  // [32-tile][r1]
+  unsigned wB[2][4];
 };
 
-DEV_INLINE CShuffled cshuffle(const unsigned pw[4][2], int g, int c) {
-  const int L0 = 32 * (g & 1) + c;
-  CShuffled s;
-#pragma unroll
-  for (int qt = 0; qt < 4; ++qt)
-#pragma unroll
-    for (int wd = 0; wd < 2; ++wd) {
-      s.sh[qt][wd][0] = __shfl(pw[qt][wd], L0, WAVE);
-      s.sh[qt][wd][1] = __shfl(pw[qt][wd], L0 + 16, WAVE);
-    }
-  return s;
-}
-
-template <int KS>
-DEV_INLINE bfrag frag_from_shuffled(const CShuffled& s, int g) {
-  const bool hi = (g >> 1) != 0;
+template <int S>
+DEV_INLINE bfrag frag_from_packed(const PackedC& p) {
+  const unsigned uA = p.wA[S >> 1][2 * (S & 1)];
+  const unsigned vA = p.wA[S >> 1][2 * (S & 1) + 1];
+  const unsigned uB = p.wB[S >> 1][2 * (S & 1)];
+  const unsigned vB = p.wB[S >> 1][2 * (S & 1) + 1];
+  auto rA = __builtin_amdgcn_permlane32_swap(uA, vA, false, false);
+  auto rB = __builtin_amdgcn_permlane32_swap(uB, vB, false, false);
   union { bfrag f; unsigned w[4]; } r;
-  r.w[0] = hi ? s.sh[2 * KS + 1][0][0] : s.sh[2 * KS][0][0];
-  r.w[1] = hi ? s.sh[2 * KS + 1][1][0] : s.sh[2 * KS][1][0];
-  r.w[2] = hi ? s.sh[2 * KS + 1][0][1] : s.sh[2 * KS][0][1];
-  r.w[3] = hi ? s.sh[2 * KS + 1][1][1] : s.sh[2 * KS][1][1];
+  r.w[0] = rA[0];
+  r.w[1] = rB[0];
+  r.w[2] = rA[1];
+  r.w[3] = rB[1];
   return r.f;
 }
 
-// load a global B/A-style fragment (row r0+c, 8 elems at col c0+8g) scaled
 DEV_INLINE bfrag load_frag_scaled(const bf16* p, int row, int col, int st,
                                   float s) {
   short8v v = load8(p + (long long)row * st + col);
@@ -171,14 +171,14 @@ DEV_INLINE bfrag load_frag(const bf16* p, int row, int col, int st) {
 // Forward
 // ---------------------------------------------------------------------------
 template <int NW>
-__launch_bounds__(NW * WAVE, 2)
+__launch_bounds__(NW * WAVE)
 __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
                                 const bf16* __restrict__ k,
                                 const bf16* __restrict__ v,
                                 bf16* __restrict__ o, float* __restrict__ lse,
                                 int T, int H, float scale, GStride sq,
                                 GStride so) {
-  constexpr int BM = NW * 16;  // q rows per workgroup
+  constexpr int BM = NW * 32;
   constexpr int NT = NW * WAVE;
   __shared__ __attribute__((aligned(16))) char smem[2 * KVB * D * 2];
   char* lds_k = smem;                 // [64][64] keys row-major
@@ -189,40 +189,39 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int w = tid / WAVE;
-  const int g = lane >> 4;
-  const int c = lane & 15;
+  const int q32 = lane & 31;
+  const int h32 = lane >> 5;
 
   const long long boff = (bh / H) * sq.b + (bh % H) * sq.h;
   const bf16* qp = q + boff + (long long)(qb * BM) * sq.t;
   const bf16* kp = k + boff;
   const bf16* vp = v + boff;
 
-  // addressing, hoisted out of the K/V loop
   const TileAddr ta = tile_addr<NT>(tid, sq.t);
-  int tdst[2][8];
+  int tdst[4][8];
   transposed_dst<NT>(tid, tdst);
-  int kf_off[4][2], vf_off[4][2];
+  int kf_off[2][4], vf_off[2][4];  // [32-tile][k-slice]
 #pragma unroll
-  for (int t16 = 0; t16 < 4; ++t16)
+  for (int t2 = 0; t2 < 2; ++t2)
 #pragma unroll
-    for (int ds = 0; ds < 2; ++ds) {
-      kf_off[t16][ds] = swz(t16 * 16 + c, (ds * 32 + 8 * g) * 2);
-      vf_off[t16][ds] = kf_off[t16][ds];  // same pattern on the V^T image
+    for (int s = 0; s < 4; ++s) {
+      kf_off[t2][s] = swz(t2 * 32 + q32, (s * 16 + 8 * h32) * 2);
+      vf_off[t2][s] = kf_off[t2][s];
     }
 
-  // Q fragments, prescaled by scale*log2e (softmax runs in exp2 space)
-  const float qs = scale * LOG2E;
-  bfrag q_frag[2];
+  const float qscale = scale * LOG2E;
+  bfrag q_frag[4];
 #pragma unroll
-  for (int ds = 0; ds < 2; ++ds)
-    q_frag[ds] = load_frag_scaled(qp, w * 16 + c, ds * 32 + 8 * g, sq.t, qs);
+  for (int s = 0; s < 4; ++s)
+    q_frag[s] = load_frag_scaled(qp, w * 32 + q32, s * 16 + 8 * h32, sq.t,
+                                 qscale);
 
-  f32x4 o_acc[4] = {};
+  f32x16 o_acc[2] = {};
   float m_run = -INFINITY;
   float l_run = 0.f;
 
-  const int row_lo = qb * BM + w * 16;       // this wave's first q row
-  const int row_me = row_lo + c;             // this lane's q row
+  const int row_lo = qb * BM + w * 32;
+  const int row_me = row_lo + q32;
   const int n_kv = (qb + 1) * BM / KVB;
   for (int j = 0; j < n_kv; ++j) {
     __syncthreads();
@@ -231,96 +230,99 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
     __syncthreads();
 
     const int key0 = j * KVB;
-    if (key0 > row_lo + 15) continue;        // wave fully above this tile
+    if (key0 > row_lo + 31) continue;
 
-    // S^T tiles: C[key = 16*sub + 4g + r][qrow = c]  (values are log2-scaled)
-    f32x4 st[4];
+    // S^T tiles: C[key = 32*t2 + crow(r,h32)][qrow = q32], log2-scaled
+    f32x16 st[2];
 #pragma unroll
-    for (int sub = 0; sub < 4; ++sub) {
-      f32x4 acc = {};
-      acc = MFMA_BF16_16x16x32(lds_read16(lds_k, kf_off[sub][0]), q_frag[0], acc);
-      acc = MFMA_BF16_16x16x32(lds_read16(lds_k, kf_off[sub][1]), q_frag[1], acc);
-      st[sub] = acc;
+    for (int t2 = 0; t2 < 2; ++t2) {
+      f32x16 acc = {};
+#pragma unroll
+      for (int s = 0; s < 4; ++s)
+        acc = MFMA32(lds_read16(lds_k, kf_off[t2][s]), q_frag[s], acc);
+      st[t2] = acc;
     }
 
-    const bool diag = key0 + KVB - 1 > row_lo;  // some key may exceed a row
+    const bool diag = key0 + KVB - 1 > row_lo;
     float mt = -INFINITY;
     if (diag) {
 #pragma unroll
-      for (int sub = 0; sub < 4; ++sub)
+      for (int t2 = 0; t2 < 2; ++t2)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int key = key0 + sub * 16 + 4 * g + r;
-          float s = (key <= row_me) ? st[sub][r] : -INFINITY;
-          st[sub][r] = s;
+        for (int r = 0; r < 16; ++r) {
+          const int key = key0 + t2 * 32 + crow(r, h32);
+          float s = (key <= row_me) ? st[t2][r] : -INFINITY;
+          st[t2][r] = s;
           mt = fmaxf(mt, s);
         }
     } else {
 #pragma unroll
-      for (int sub = 0; sub < 4; ++sub)
+      for (int t2 = 0; t2 < 2; ++t2)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) mt = fmaxf(mt, st[sub][r]);
+        for (int r = 0; r < 16; ++r) mt = fmaxf(mt, st[t2][r]);
     }
-    mt = fmaxf(mt, __shfl_xor(mt, 16, WAVE));
     mt = fmaxf(mt, __shfl_xor(mt, 32, WAVE));
     const float m_new = fmaxf(m_run, mt);
     const float alpha = exp2f(m_run - m_new);
     float psum = 0.f;
-    unsigned pw[4][2];
+    PackedC P;
 #pragma unroll
-    for (int sub = 0; sub < 4; ++sub) {
-      // masked scores are -inf and m_new is finite, so exp2 gives exact 0
-      const float p0 = exp2f(st[sub][0] - m_new);
-      const float p1 = exp2f(st[sub][1] - m_new);
-      const float p2 = exp2f(st[sub][2] - m_new);
-      const float p3 = exp2f(st[sub][3] - m_new);
-      psum += p0 + p1 + p2 + p3;
-      pw[sub][0] = pack2(p0, p1);
-      pw[sub][1] = pack2(p2, p3);
-    }
-    psum += __shfl_xor(psum, 16, WAVE);
+    for (int t2 = 0; t2 < 2; ++t2)
+#pragma unroll
+      for (int r1 = 0; r1 < 4; ++r1) {
+        const float p0 = exp2f(st[t2][4 * r1 + 0] - m_new);
+        const float p1 = exp2f(st[t2][4 * r1 + 1] - m_new);
+        const float p2 = exp2f(st[t2][4 * r1 + 2] - m_new);
+        const float p3 = exp2f(st[t2][4 * r1 + 3] - m_new);
+        psum += p0 + p1 + p2 + p3;
+        P.wA[t2][r1] = pack2(p0, p1);
+        P.wB[t2][r1] = pack2(p2, p3);
+      }
     psum += __shfl_xor(psum, 32, WAVE);
     l_run = l_run * alpha + psum;
     m_run = m_new;
 
-    // O rescale (skip when alpha == 1 for every row of the wave)
     if (__any(alpha != 1.0f)) {
-      float alpha_row[4];
+      float a_row[16];
 #pragma unroll
-      for (int r = 0; r < 4; ++r) alpha_row[r] = __shfl(alpha, 4 * g + r, WAVE);
+      for (int r = 0; r < 16; ++r)
+        a_row[r] = __shfl(alpha, crow(r, h32), WAVE);
 #pragma unroll
-      for (int dt = 0; dt < 4; ++dt)
+      for (int dt = 0; dt < 2; ++dt)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) o_acc[dt][r] *= alpha_row[r];
+        for (int r = 0; r < 16; ++r) o_acc[dt][r] *= a_row[r];
     }
 
     // PV
-    const CShuffled psh = cshuffle(pw, g, c);
-    const bfrag pa0 = frag_from_shuffled<0>(psh, g);
-    const bfrag pa1 = frag_from_shuffled<1>(psh, g);
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
-      o_acc[dt] = MFMA_BF16_16x16x32(pa0, lds_read16(lds_vt, vf_off[dt][0]),
-                                     o_acc[dt]);
-      o_acc[dt] = MFMA_BF16_16x16x32(pa1, lds_read16(lds_vt, vf_off[dt][1]),
-                                     o_acc[dt]);
+    for (int s = 0; s < 4; ++s) {
+      bfrag pa;
+      switch (s) {
+        case 0: pa = frag_from_packed<0>(P); break;
+        case 1: pa = frag_from_packed<1>(P); break;
+        case 2: pa = frag_from_packed<2>(P); break;
+        default: pa = frag_from_packed<3>(P); break;
+      }
+#pragma unroll
+      for (int dt = 0; dt < 2; ++dt)
+        o_acc[dt] = MFMA32(pa, lds_read16(lds_vt, vf_off[dt][s]), o_acc[dt]);
     }
   }
 
-  // epilogue: O /= l ; lse = (m + log2(l)) * ln2
-  float linv_row[4];
+  float linv_row[16];
 #pragma unroll
-  for (int r = 0; r < 4; ++r)
-    linv_row[r] = 1.0f / __shfl(l_run, 4 * g + r, WAVE);
+  for (int r = 0; r < 16; ++r)
+    linv_row[r] = 1.0f / __shfl(l_run, crow(r, h32), WAVE);
   bf16* op = o + (bh / H) * so.b + (bh % H) * so.h
-             + (long long)(qb * BM + w * 16) * so.t;
+             + (long long)(qb * BM + w * 32) * so.t;
 #pragma unroll
-  for (int dt = 0; dt < 4; ++dt)
+  for (int dt = 0; dt < 2; ++dt)
 #pragma unroll
-    for (int r = 0; r < 4; ++r)
-      op[(4 * g + r) * so.t + dt * 16 + c] = f2bf(o_acc[dt][r] * linv_row[r]);
-  if (lane < 16) {
-    lse[bh * T + qb * BM + w * 16 + c] = (m_run + log2f(l_run)) * LN2;
+    for (int r = 0; r < 16; ++r)
+      op[crow(r, h32) * so.t + dt * 32 + q32] =
+          f2bf(o_acc[dt][r] * linv_row[r]);
+  if (lane < 32) {
+    lse[bh * T + qb * BM + w * 32 + q32] = (m_run + log2f(l_run)) * LN2;
   }
 }
 
@@ -344,10 +346,10 @@ __global__ void attn_delta_kernel(const bf16* __restrict__ dout,
 }
 
 // ---------------------------------------------------------------------------
-// Backward dK/dV: one workgroup per (NW*16)-key block; wave w owns 16 keys.
+// Backward dK/dV: one workgroup per (NW*32)-key block; wave w owns 32 keys.
 // ---------------------------------------------------------------------------
 template <int NW>
-__launch_bounds__(NW * WAVE, 2)
+__launch_bounds__(NW * WAVE)
 __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
                                     const bf16* __restrict__ k,
                                     const bf16* __restrict__ v,
@@ -357,7 +359,7 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
                                     bf16* __restrict__ dk, bf16* __restrict__ dv,
                                     int T, int H, float scale, GStride sq,
                                     GStride so, GStride sd) {
-  constexpr int BK = NW * 16;  // keys per workgroup
+  constexpr int BK = NW * 32;
   constexpr int NT = NW * WAVE;
   __shared__ __attribute__((aligned(16))) char smem[4 * KVB * D * 2 + 2 * KVB * 4];
   char* lds_q = smem;
@@ -372,8 +374,8 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int w = tid / WAVE;
-  const int g = lane >> 4;
-  const int c = lane & 15;
+  const int k32 = lane & 31;   // this lane's key within the wave tile
+  const int h32 = lane >> 5;
 
   const long long boff = (bh / H) * sq.b + (bh % H) * sq.h;
   const long long ooff = (bh / H) * so.b + (bh % H) * so.h;
@@ -384,31 +386,30 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
 
   const TileAddr ta = tile_addr<NT>(tid, sq.t);
   const TileAddr tao = tile_addr<NT>(tid, so.t);
-  int tdst[2][8];
+  int tdst[4][8];
   transposed_dst<NT>(tid, tdst);
-  int af_off[4][2], bf_off[4][2];
+  int af_off[2][4], bf_off[2][4];
 #pragma unroll
-  for (int t16 = 0; t16 < 4; ++t16)
+  for (int t2 = 0; t2 < 2; ++t2)
 #pragma unroll
-    for (int ds = 0; ds < 2; ++ds) {
-      af_off[t16][ds] = swz(t16 * 16 + c, (ds * 32 + 8 * g) * 2);
-      bf_off[t16][ds] = af_off[t16][ds];
+    for (int s = 0; s < 4; ++s) {
+      af_off[t2][s] = swz(t2 * 32 + k32, (s * 16 + 8 * h32) * 2);
+      bf_off[t2][s] = af_off[t2][s];
     }
 
-  // This wave's K (prescaled by scale*log2e) and V fragments in registers.
-  const float ks_scale = scale * LOG2E;
-  const int key_lo = jb * BK + w * 16;   // first key of this wave
-  const int key_me = key_lo + c;         // this lane's key
-  bfrag k_frag[2], v_frag[2];
+  const float kscale = scale * LOG2E;
+  const int key_lo = jb * BK + w * 32;
+  const int key_me = key_lo + k32;
+  bfrag k_frag[4], v_frag[4];  // B operands: col = k32, k-slices over D
 #pragma unroll
-  for (int ds = 0; ds < 2; ++ds) {
-    k_frag[ds] = load_frag_scaled(kp, jb * BK + w * 16 + c, ds * 32 + 8 * g,
-                                  sq.t, ks_scale);
-    v_frag[ds] = load_frag(vp, jb * BK + w * 16 + c, ds * 32 + 8 * g, sq.t);
+  for (int s = 0; s < 4; ++s) {
+    k_frag[s] = load_frag_scaled(kp, key_lo + k32, s * 16 + 8 * h32, sq.t,
+                                 kscale);
+    v_frag[s] = load_frag(vp, key_lo + k32, s * 16 + 8 * h32, sq.t);
   }
 
-  f32x4 dk_acc[4] = {};
-  f32x4 dv_acc[4] = {};
+  f32x16 dk_acc[2] = {};
+  f32x16 dv_acc[2] = {};
 
   for (int i = jb * BK / KVB; i < T / KVB; ++i) {
     __syncthreads();
@@ -423,69 +424,77 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
     __syncthreads();
 
     const int q0 = i * KVB;
-    if (q0 + KVB - 1 < key_lo) continue;   // all rows above this wave's keys
+    if (q0 + KVB - 1 < key_lo) continue;
 
-    const bool diag = q0 < key_lo + 16;    // some row may precede a key
-    unsigned pwp[4][2];
-    unsigned pws[4][2];
+    const bool diag = q0 < key_lo + 32;
+    PackedC P, dS;
 #pragma unroll
-    for (int qt = 0; qt < 4; ++qt) {
-      f32x4 s_acc = {};
-      f32x4 dp_acc = {};
-      s_acc = MFMA_BF16_16x16x32(lds_read16(lds_q, af_off[qt][0]), k_frag[0], s_acc);
-      s_acc = MFMA_BF16_16x16x32(lds_read16(lds_q, af_off[qt][1]), k_frag[1], s_acc);
-      dp_acc = MFMA_BF16_16x16x32(lds_read16(lds_do, af_off[qt][0]), v_frag[0], dp_acc);
-      dp_acc = MFMA_BF16_16x16x32(lds_read16(lds_do, af_off[qt][1]), v_frag[1], dp_acc);
-      float p[4], dsv[4];
+    for (int t2 = 0; t2 < 2; ++t2) {
+      // S, dP tiles: C[qrow = 32*t2 + crow(r,h32)][key = k32]
+      f32x16 s_acc = {};
+      f32x16 dp_acc = {};
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qrow = q0 + qt * 16 + 4 * g + r;
-        const float l2 = lds_lse[qt * 16 + 4 * g + r];
-        const float dlt = lds_dlt[qt * 16 + 4 * g + r];
+      for (int s = 0; s < 4; ++s) {
+        s_acc = MFMA32(lds_read16(lds_q, af_off[t2][s]), k_frag[s], s_acc);
+        dp_acc = MFMA32(lds_read16(lds_do, bf_off[t2][s]), v_frag[s], dp_acc);
+      }
+      float p[16], dsv[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int lrow = t2 * 32 + crow(r, h32);
+        const int qrow = q0 + lrow;
+        const float l2 = lds_lse[lrow];
+        const float dlt = lds_dlt[lrow];
         float pp = exp2f(s_acc[r] - l2);
         if (diag) pp = (key_me <= qrow) ? pp : 0.f;
         p[r] = pp;
-        dsv[r] = pp * (dp_acc[r] - dlt);   // scale folded into epilogue
+        dsv[r] = pp * (dp_acc[r] - dlt);  // scale folded into epilogue
       }
-      pwp[qt][0] = pack2(p[0], p[1]);
-      pwp[qt][1] = pack2(p[2], p[3]);
-      pws[qt][0] = pack2(dsv[0], dsv[1]);
-      pws[qt][1] = pack2(dsv[2], dsv[3]);
+#pragma unroll
+      for (int r1 = 0; r1 < 4; ++r1) {
+        P.wA[t2][r1] = pack2(p[4 * r1], p[4 * r1 + 1]);
+        P.wB[t2][r1] = pack2(p[4 * r1 + 2], p[4 * r1 + 3]);
+        dS.wA[t2][r1] = pack2(dsv[4 * r1], dsv[4 * r1 + 1]);
+        dS.wB[t2][r1] = pack2(dsv[4 * r1 + 2], dsv[4 * r1 + 3]);
+      }
     }
 
-    const CShuffled pshp = cshuffle(pwp, g, c);
-    const CShuffled pshs = cshuffle(pws, g, c);
-    const bfrag ap0 = frag_from_shuffled<0>(pshp, g);
-    const bfrag ap1 = frag_from_shuffled<1>(pshp, g);
-    const bfrag as0 = frag_from_shuffled<0>(pshs, g);
-    const bfrag as1 = frag_from_shuffled<1>(pshs, g);
+    // dV[key][d] += P^T dO ; dK[key][d] += dS^T Q  (A-frag k = qrow)
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
-      dv_acc[dt] = MFMA_BF16_16x16x32(ap0, lds_read16(lds_dot, bf_off[dt][0]), dv_acc[dt]);
-      dv_acc[dt] = MFMA_BF16_16x16x32(ap1, lds_read16(lds_dot, bf_off[dt][1]), dv_acc[dt]);
-      dk_acc[dt] = MFMA_BF16_16x16x32(as0, lds_read16(lds_qt, bf_off[dt][0]), dk_acc[dt]);
-      dk_acc[dt] = MFMA_BF16_16x16x32(as1, lds_read16(lds_qt, bf_off[dt][1]), dk_acc[dt]);
+    for (int s = 0; s < 4; ++s) {
+      bfrag ap, as;
+      switch (s) {
+        case 0: ap = frag_from_packed<0>(P); as = frag_from_packed<0>(dS); break;
+        case 1: ap = frag_from_packed<1>(P); as = frag_from_packed<1>(dS); break;
+        case 2: ap = frag_from_packed<2>(P); as = frag_from_packed<2>(dS); break;
+        default: ap = frag_from_packed<3>(P); as = frag_from_packed<3>(dS); break;
+      }
+#pragma unroll
+      for (int dt = 0; dt < 2; ++dt) {
+        dv_acc[dt] = MFMA32(ap, lds_read16(lds_dot, bf_off[dt][s]), dv_acc[dt]);
+        dk_acc[dt] = MFMA32(as, lds_read16(lds_qt, bf_off[dt][s]), dk_acc[dt]);
+      }
     }
   }
 
   const long long doff = (bh / H) * sd.b + (bh % H) * sd.h
-                         + (long long)(jb * BK + w * 16) * sd.t;
+                         + (long long)(key_lo) * sd.t;
   bf16* dkp = dk + doff;
   bf16* dvp = dv + doff;
 #pragma unroll
-  for (int dt = 0; dt < 4; ++dt)
+  for (int dt = 0; dt < 2; ++dt)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      dkp[(4 * g + r) * sd.t + dt * 16 + c] = f2bf(dk_acc[dt][r] * scale);
-      dvp[(4 * g + r) * sd.t + dt * 16 + c] = f2bf(dv_acc[dt][r]);
+    for (int r = 0; r < 16; ++r) {
+      dkp[crow(r, h32) * sd.t + dt * 32 + k32] = f2bf(dk_acc[dt][r] * scale);
+      dvp[crow(r, h32) * sd.t + dt * 32 + k32] = f2bf(dv_acc[dt][r]);
     }
 }
 
 // ---------------------------------------------------------------------------
-// Backward dQ: one workgroup per (NW*16)-row Q block; wave w owns 16 rows.
+// Backward dQ: one workgroup per (NW*32)-row Q block; wave w owns 32 rows.
 // ---------------------------------------------------------------------------
 template <int NW>
-__launch_bounds__(NW * WAVE, 2)
+__launch_bounds__(NW * WAVE)
 __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
                                    const bf16* __restrict__ k,
                                    const bf16* __restrict__ v,
@@ -495,7 +504,7 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
                                    bf16* __restrict__ dq, int T, int H,
                                    float scale, GStride sq, GStride so,
                                    GStride sd) {
-  constexpr int BM = NW * 16;
+  constexpr int BM = NW * 32;
   constexpr int NT = NW * WAVE;
   __shared__ __attribute__((aligned(16))) char smem[3 * KVB * D * 2];
   char* lds_k = smem;                     // K row-major (A of S^T)
@@ -507,8 +516,8 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int w = tid / WAVE;
-  const int g = lane >> 4;
-  const int c = lane & 15;
+  const int q32 = lane & 31;
+  const int h32 = lane >> 5;
 
   const long long boff = (bh / H) * sq.b + (bh % H) * sq.h;
   const long long ooff = (bh / H) * so.b + (bh % H) * so.h;
@@ -518,28 +527,29 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   const bf16* dop = dout + ooff + (long long)(qb * BM) * so.t;
 
   const TileAddr ta = tile_addr<NT>(tid, sq.t);
-  int tdst[2][8];
+  int tdst[4][8];
   transposed_dst<NT>(tid, tdst);
-  int f_off[4][2];
+  int f_off[2][4];
 #pragma unroll
-  for (int t16 = 0; t16 < 4; ++t16)
+  for (int t2 = 0; t2 < 2; ++t2)
 #pragma unroll
-    for (int ds = 0; ds < 2; ++ds)
-      f_off[t16][ds] = swz(t16 * 16 + c, (ds * 32 + 8 * g) * 2);
+    for (int s = 0; s < 4; ++s)
+      f_off[t2][s] = swz(t2 * 32 + q32, (s * 16 + 8 * h32) * 2);
 
-  const float qs = scale * LOG2E;
-  bfrag q_frag[2], do_frag[2];
+  const float qscale = scale * LOG2E;
+  bfrag q_frag[4], do_frag[4];
 #pragma unroll
-  for (int ds = 0; ds < 2; ++ds) {
-    q_frag[ds] = load_frag_scaled(qp, w * 16 + c, ds * 32 + 8 * g, sq.t, qs);
-    do_frag[ds] = load_frag(dop, w * 16 + c, ds * 32 + 8 * g, so.t);
+  for (int s = 0; s < 4; ++s) {
+    q_frag[s] = load_frag_scaled(qp, w * 32 + q32, s * 16 + 8 * h32, sq.t,
+                                 qscale);
+    do_frag[s] = load_frag(dop, w * 32 + q32, s * 16 + 8 * h32, so.t);
   }
-  const int row_lo = qb * BM + w * 16;
-  const int row_me = row_lo + c;
+  const int row_lo = qb * BM + w * 32;
+  const int row_me = row_lo + q32;
   const float lse2_me = lse[bh * T + row_me] * LOG2E;
   const float dlt_me = delta[bh * T + row_me];
 
-  f32x4 dq_acc[4] = {};
+  f32x16 dq_acc[2] = {};
 
   const int n_kv = (qb + 1) * BM / KVB;
   for (int j = 0; j < n_kv; ++j) {
@@ -550,47 +560,58 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
     __syncthreads();
 
     const int key0 = j * KVB;
-    if (key0 > row_lo + 15) continue;
+    if (key0 > row_lo + 31) continue;
 
     const bool diag = key0 + KVB - 1 > row_lo;
-    unsigned pws[4][2];
+    PackedC dS;
 #pragma unroll
-    for (int sub = 0; sub < 4; ++sub) {
-      f32x4 s_acc = {};
-      f32x4 dp_acc = {};
-      s_acc = MFMA_BF16_16x16x32(lds_read16(lds_k, f_off[sub][0]), q_frag[0], s_acc);
-      s_acc = MFMA_BF16_16x16x32(lds_read16(lds_k, f_off[sub][1]), q_frag[1], s_acc);
-      dp_acc = MFMA_BF16_16x16x32(lds_read16(lds_v, f_off[sub][0]), do_frag[0], dp_acc);
-      dp_acc = MFMA_BF16_16x16x32(lds_read16(lds_v, f_off[sub][1]), do_frag[1], dp_acc);
-      float dsv[4];
+    for (int t2 = 0; t2 < 2; ++t2) {
+      // S^T, dP^T tiles: C[key = 32*t2 + crow(r,h32)][qrow = q32]
+      f32x16 s_acc = {};
+      f32x16 dp_acc = {};
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int key = key0 + sub * 16 + 4 * g + r;
+      for (int s = 0; s < 4; ++s) {
+        s_acc = MFMA32(lds_read16(lds_k, f_off[t2][s]), q_frag[s], s_acc);
+        dp_acc = MFMA32(lds_read16(lds_v, f_off[t2][s]), do_frag[s], dp_acc);
+      }
+      float dsv[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int key = key0 + t2 * 32 + crow(r, h32);
         float pp = exp2f(s_acc[r] - lse2_me);
         if (diag) pp = (key <= row_me) ? pp : 0.f;
-        dsv[r] = pp * (dp_acc[r] - dlt_me);  // scale folded into epilogue
+        dsv[r] = pp * (dp_acc[r] - dlt_me);
       }
-      pws[sub][0] = pack2(dsv[0], dsv[1]);
-      pws[sub][1] = pack2(dsv[2], dsv[3]);
+#pragma unroll
+      for (int r1 = 0; r1 < 4; ++r1) {
+        dS.wA[t2][r1] = pack2(dsv[4 * r1], dsv[4 * r1 + 1]);
+        dS.wB[t2][r1] = pack2(dsv[4 * r1 + 2], dsv[4 * r1 + 3]);
+      }
     }
 
-    const CShuffled pshs = cshuffle(pws, g, c);
-    const bfrag as0 = frag_from_shuffled<0>(pshs, g);
-    const bfrag as1 = frag_from_shuffled<1>(pshs, g);
+    // dQ[qrow][d] += dS K  (A-frag k = key)
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
-      dq_acc[dt] = MFMA_BF16_16x16x32(as0, lds_read16(lds_kt, f_off[dt][0]), dq_acc[dt]);
-      dq_acc[dt] = MFMA_BF16_16x16x32(as1, lds_read16(lds_kt, f_off[dt][1]), dq_acc[dt]);
+    for (int s = 0; s < 4; ++s) {
+      bfrag as;
+      switch (s) {
+        case 0: as = frag_from_packed<0>(dS); break;
+        case 1: as = frag_from_packed<1>(dS); break;
+        case 2: as = frag_from_packed<2>(dS); break;
+        default: as = frag_from_packed<3>(dS); break;
+      }
+#pragma unroll
+      for (int dt = 0; dt < 2; ++dt)
+        dq_acc[dt] = MFMA32(as, lds_read16(lds_kt, f_off[dt][s]), dq_acc[dt]);
     }
   }
 
   bf16* dqp = dq + (bh / H) * sd.b + (bh % H) * sd.h
-              + (long long)(qb * BM + w * 16) * sd.t;
+              + (long long)(qb * BM + w * 32) * sd.t;
 #pragma unroll
-  for (int dt = 0; dt < 4; ++dt)
+  for (int dt = 0; dt < 2; ++dt)
 #pragma unroll
-    for (int r = 0; r < 4; ++r)
-      dqp[(4 * g + r) * sd.t + dt * 16 + c] = f2bf(dq_acc[dt][r] * scale);
+    for (int r = 0; r < 16; ++r)
+      dqp[crow(r, h32) * sd.t + dt * 32 + q32] = f2bf(dq_acc[dt][r] * scale);
 }
 
 }  // namespace
@@ -606,17 +627,15 @@ hipError_t tdsa_attn_fwd(const void* q, const void* k, const void* v, void* o,
   if (T % KVB) return hipErrorInvalidValue;
   GStride sq{sq_in[0], sq_in[1], (int)sq_in[2]};
   GStride so{so_in[0], so_in[1], (int)so_in[2]};
-  if (T % 128 == 0) {
-    dim3 grid(T / 128, B * H);
-    hipLaunchKernelGGL(attn_fwd_kernel<8>, grid, dim3(512), 0, stream,
-                       (const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o,
-                       lse, T, (int)H, scale, sq, so);
-  } else {
-    dim3 grid(T / 64, B * H);
-    hipLaunchKernelGGL(attn_fwd_kernel<4>, grid, dim3(256), 0, stream,
-                       (const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o,
-                       lse, T, (int)H, scale, sq, so);
-  }
+#define LAUNCH_FWD(NW)                                                        \
+  hipLaunchKernelGGL(attn_fwd_kernel<NW>, dim3(T / (NW * 32), B * H),         \
+                     dim3(NW * WAVE), 0, stream, (const bf16*)q,              \
+                     (const bf16*)k, (const bf16*)v, (bf16*)o, lse, T,        \
+                     (int)H, scale, sq, so)
+  if (T % 256 == 0) LAUNCH_FWD(8);
+  else if (T % 128 == 0) LAUNCH_FWD(4);
+  else LAUNCH_FWD(2);
+#undef LAUNCH_FWD
   return hipGetLastError();
 }
 
@@ -639,27 +658,22 @@ hipError_t tdsa_attn_bwd(const void* q, const void* k, const void* v,
                        (const bf16*)dout, (const bf16*)o, delta, R, T, (int)H,
                        so);
   }
-  if (T % 128 == 0) {
-    dim3 grid(T / 128, BH);
-    hipLaunchKernelGGL(attn_bwd_dkv_kernel<8>, grid, dim3(512), 0, stream,
-                       (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                       (const bf16*)dout, lse, delta, (bf16*)dk, (bf16*)dv, T,
-                       (int)H, scale, sq, so, sd);
-    hipLaunchKernelGGL(attn_bwd_dq_kernel<8>, grid, dim3(512), 0, stream,
-                       (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                       (const bf16*)dout, lse, delta, (bf16*)dq, T, (int)H,
-                       scale, sq, so, sd);
-  } else {
-    dim3 grid(T / 64, BH);
-    hipLaunchKernelGGL(attn_bwd_dkv_kernel<4>, grid, dim3(256), 0, stream,
-                       (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                       (const bf16*)dout, lse, delta, (bf16*)dk, (bf16*)dv, T,
-                       (int)H, scale, sq, so, sd);
-    hipLaunchKernelGGL(attn_bwd_dq_kernel<4>, grid, dim3(256), 0, stream,
-                       (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                       (const bf16*)dout, lse, delta, (bf16*)dq, T, (int)H,
-                       scale, sq, so, sd);
-  }
+#define LAUNCH_BWD(NW)                                                        \
+  do {                                                                        \
+    dim3 grid(T / (NW * 32), BH);                                             \
+    hipLaunchKernelGGL(attn_bwd_dkv_kernel<NW>, grid, dim3(NW * WAVE), 0,     \
+                       stream, (const bf16*)q, (const bf16*)k,                \
+                       (const bf16*)v, (const bf16*)dout, lse, delta,         \
+                       (bf16*)dk, (bf16*)dv, T, (int)H, scale, sq, so, sd);   \
+    hipLaunchKernelGGL(attn_bwd_dq_kernel<NW>, grid, dim3(NW * WAVE), 0,      \
+                       stream, (const bf16*)q, (const bf16*)k,                \
+                       (const bf16*)v, (const bf16*)dout, lse, delta,         \
+                       (bf16*)dq, T, (int)H, scale, sq, so, sd);              \
+  } while (0)
+  if (T % 256 == 0) LAUNCH_BWD(8);
+  else if (T % 128 == 0) LAUNCH_BWD(4);
+  else LAUNCH_BWD(2);
+#undef LAUNCH_BWD
   return hipGetLastError();
 }
 

@@ -178,9 +178,10 @@ def _attn_ref(q, k, v, scale):
     return torch.softmax(s, dim=-1) @ vf
 
 
-def test_attention_fwd_gpu():
+@pytest.mark.parametrize("T", [128, 192, 256])  # covers NW=4, NW=2, NW=8
+def test_attention_fwd_gpu(T):
     torch.manual_seed(0)
-    B, H, T, D = 2, 3, 256, 64
+    B, H, D = 2, 3, 64
     q = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16)
     k = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16)
     v = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16)
@@ -196,9 +197,10 @@ def test_attention_fwd_gpu():
     _close(lse, lse_ref, 2e-2, "attn lse")
 
 
-def test_attention_bwd_gpu():
+@pytest.mark.parametrize("T", [128, 192, 256])
+def test_attention_bwd_gpu(T):
     torch.manual_seed(1)
-    B, H, T, D = 2, 2, 256, 64
+    B, H, D = 2, 2, 64
     scale = 1.0 / math.sqrt(D)
     q = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16,
                     requires_grad=True)
