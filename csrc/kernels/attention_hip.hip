@@ -115,6 +115,13 @@ DEV_INLINE void stage_transposed(const bf16* __restrict__ g, char* lds,
   }
 }
 
+// Transposed images use a second XOR term (row bits 3-5): during the scalar
+// write burst the dest row is congruent mod 8 across a half-wave, so the
+// plain swizzle degenerates to f(source-row-only) banks (16-way measured,
+// 13.5% of dkv wave cycles); row bits 3-5 vary across the burst and spread
+// it to <=2-way. Readers XOR the same term (swzT(r,b) = swz(r,b) ^ trx(r)).
+DEV_INLINE int trx(int row) { return ((row >> 3) & 7) << 4; }
+
 template <int NT>
 DEV_INLINE void transposed_dst(int tid, int (&tdst)[4][8]) {
 #pragma unroll
@@ -123,7 +130,8 @@ DEV_INLINE void transposed_dst(int tid, int (&tdst)[4][8]) {
     int row = chunk >> 3;
     int c0 = (chunk & 7) * 8;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) tdst[rep][j] = swz(c0 + j, row * 2);
+    for (int j = 0; j < 8; ++j)
+      tdst[rep][j] = swz(c0 + j, row * 2) ^ trx(c0 + j);
   }
 }
 
@@ -211,6 +219,7 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
     for (int s = 0; s < 4; ++s)
       kf_off[t2][s] = swz(t2 * 32 + q32, (s * 16 + 8 * h32) * 2);
 
+  const int xtr[2] = {trx(q32), trx(32 + q32)};
   const float qscale = scale * LOG2E;
   bfrag q_frag[4];
 #pragma unroll
@@ -307,7 +316,8 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
       }
 #pragma unroll
       for (int dt = 0; dt < 2; ++dt)
-        o_acc[dt] = MFMA32(pa, lds_read16(lds_vt, kf_off[dt][s]), o_acc[dt]);
+        o_acc[dt] = MFMA32(pa, lds_read16(lds_vt, kf_off[dt][s] ^ xtr[dt]),
+                           o_acc[dt]);
     }
   }
 
@@ -331,20 +341,21 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
 // ---------------------------------------------------------------------------
 // Backward: delta = rowsum(dO * O)
 // ---------------------------------------------------------------------------
+// grid: (ceil(T/waves_per_block), B*H) — no per-thread 64-bit div/mod (a
+// 1-D flat-index version spent ~234 VALU/row on the divisions).
 __global__ void attn_delta_kernel(const bf16* __restrict__ dout,
                                   const bf16* __restrict__ o,
-                                  float* __restrict__ delta, long long R,
-                                  int T, int H, GStride so) {
-  const long long row = (long long)blockIdx.x * (blockDim.x / WAVE)
-                        + threadIdx.x / WAVE;
-  if (row >= R) return;
+                                  float* __restrict__ delta, int T, int H,
+                                  GStride so) {
+  const int t = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+  if (t >= T) return;
+  const long long bh = blockIdx.y;
   const int lane = threadIdx.x & (WAVE - 1);
-  const long long bh = row / T;
   const long long off = (bh / H) * so.b + (bh % H) * so.h
-                        + (long long)(row % T) * so.t + lane;
+                        + (long long)t * so.t + lane;
   float acc = bf2f(dout[off]) * bf2f(o[off]);
   acc = wave_sum(acc);
-  if (lane == 0) delta[row] = acc;
+  if (lane == 0) delta[bh * T + t] = acc;
 }
 
 // ---------------------------------------------------------------------------
@@ -397,6 +408,7 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
     for (int s = 0; s < 4; ++s)
       af_off[t2][s] = swz(t2 * 32 + k32, (s * 16 + 8 * h32) * 2);
 
+  const int xtr[2] = {trx(k32), trx(32 + k32)};
   const float kscale = scale * LOG2E;
   const int key_lo = jb * BK + w * 32;
   const int key_me = key_lo + k32;
@@ -470,8 +482,10 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
       }
 #pragma unroll
       for (int dt = 0; dt < 2; ++dt) {
-        dv_acc[dt] = MFMA32(ap, lds_read16(lds_dot, af_off[dt][s]), dv_acc[dt]);
-        dk_acc[dt] = MFMA32(as, lds_read16(lds_qt, af_off[dt][s]), dk_acc[dt]);
+        dv_acc[dt] = MFMA32(ap, lds_read16(lds_dot, af_off[dt][s] ^ xtr[dt]),
+                            dv_acc[dt]);
+        dk_acc[dt] = MFMA32(as, lds_read16(lds_qt, af_off[dt][s] ^ xtr[dt]),
+                            dk_acc[dt]);
       }
     }
   }
@@ -535,6 +549,7 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
     for (int s = 0; s < 4; ++s)
       f_off[t2][s] = swz(t2 * 32 + q32, (s * 16 + 8 * h32) * 2);
 
+  const int xtr[2] = {trx(q32), trx(32 + q32)};
   const float qscale = scale * LOG2E;
   bfrag q_frag[4], do_frag[4];
 #pragma unroll
@@ -600,7 +615,8 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
       }
 #pragma unroll
       for (int dt = 0; dt < 2; ++dt)
-        dq_acc[dt] = MFMA32(as, lds_read16(lds_kt, f_off[dt][s]), dq_acc[dt]);
+        dq_acc[dt] = MFMA32(as, lds_read16(lds_kt, f_off[dt][s] ^ xtr[dt]),
+                            dq_acc[dt]);
     }
   }
 
@@ -649,13 +665,12 @@ hipError_t tdsa_attn_bwd(const void* q, const void* k, const void* v,
   GStride so{so_in[0], so_in[1], (int)so_in[2]};
   GStride sd{sd_in[0], sd_in[1], (int)sd_in[2]};
   const long long BH = B * H;
-  const long long R = BH * T;
   {
     const int rows_per_block = 256 / WAVE;
-    const long long grid = (R + rows_per_block - 1) / rows_per_block;
-    hipLaunchKernelGGL(attn_delta_kernel, dim3(grid), dim3(256), 0, stream,
-                       (const bf16*)dout, (const bf16*)o, delta, R, T, (int)H,
-                       so);
+    hipLaunchKernelGGL(attn_delta_kernel,
+                       dim3((T + rows_per_block - 1) / rows_per_block, BH),
+                       dim3(256), 0, stream, (const bf16*)dout, (const bf16*)o,
+                       delta, T, (int)H, so);
   }
 #define LAUNCH_BWD(NW)                                                        \
   do {                                                                        \
