@@ -25,11 +25,15 @@ template <> struct RowVec<float, 4> {
   typedef float4v V;
   static DEV_INLINE V load(const float* p) { return *reinterpret_cast<const V*>(p); }
   static DEV_INLINE float get(V v, int i) { return v[i]; }
+  static DEV_INLINE void set(V& v, int i, float x) { v[i] = x; }
+  static DEV_INLINE void store(float* p, V v) { *reinterpret_cast<V*>(p) = v; }
 };
 template <> struct RowVec<bf16, 8> {
   typedef short8v V;
   static DEV_INLINE V load(const bf16* p) { return load8(p); }
   static DEV_INLINE float get(V v, int i) { return bf_elem(v, i); }
+  static DEV_INLINE void set(V& v, int i, float x) { v[i] = bf_pack(x); }
+  static DEV_INLINE void store(bf16* p, V v) { store8(p, v); }
 };
 
 // One workgroup per row (grid-strided). Writes lse[row] (fp32) and
@@ -122,6 +126,7 @@ __global__ void ce_bwd_kernel(const T* __restrict__ logits,
     const float l = lse[row];
     for (int i = threadIdx.x * W; i + W <= V; i += blockDim.x * W) {
       typename RV::V v = RV::load(lr + i);
+      typename RV::V out;
 #pragma unroll
       for (int k = 0; k < W; ++k) {
         float g = 0.f;
@@ -130,8 +135,9 @@ __global__ void ce_bwd_kernel(const T* __restrict__ logits,
           if ((long long)(i + k) == t) g -= 1.0f;
           g *= scale;
         }
-        dr[i + k] = (T)g;
+        RV::set(out, k, g);
       }
+      RV::store(dr + i, out);
     }
     // tail
     int rem0 = (V / W) * W;
