@@ -8,6 +8,8 @@
 #include <hip/hip_runtime_api.h>
 
 #include <array>
+#include <cstring>
+#include <vector>
 
 // ---- extern "C" launcher prototypes (csrc/kernels/*.hip) ------------------
 extern "C" {
@@ -35,6 +37,10 @@ hipError_t tdsa_ce_bwd(const void*, const long long*, const float*, void*,
 hipError_t tdsa_adamw_step(void*, const void*, float*, float*, float*, float*,
                            int, int, float, float, float, float, float,
                            long long, long long, int, int, hipStream_t);
+hipError_t tdsa_adamw_multi(const void*, long long, int, int, float, float,
+                            float, float, float, long long, hipStream_t);
+int tdsa_adamw_desc_size();
+int tdsa_adamw_chunkref_size();
 hipError_t tdsa_sgd_step(void*, const void*, float*, float*, int, int, float,
                          float, float, float, int, int, int, long long, int,
                          int, hipStream_t);
@@ -239,6 +245,59 @@ void adamw_step(at::Tensor param, at::Tensor grad, at::Tensor m, at::Tensor v,
             "adamw_step");
 }
 
+// One fused launch for a whole parameter list (no amsgrad; the python
+// optimizer falls back to per-tensor calls for that).
+void adamw_step_multi(std::vector<at::Tensor> params,
+                      std::vector<at::Tensor> grads,
+                      std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+                      std::vector<c10::optional<at::Tensor>> masters,
+                      double lr, double b1, double b2, double eps, double wd,
+                      int64_t step) {
+  const size_t n = params.size();
+  TORCH_CHECK(grads.size() == n && ms.size() == n && vs.size() == n &&
+              masters.size() == n, "length mismatch");
+  if (n == 0) return;
+  struct Desc {  // must mirror AdamTensorDesc in optim.hip
+    void* p; const void* g; float* m; float* v; float* master;
+    long long numel; int param_bf16; int grad_bf16;
+  };
+  struct CRef { int tensor; int chunk; };
+  TORCH_CHECK((int)sizeof(Desc) == tdsa_adamw_desc_size());
+  TORCH_CHECK((int)sizeof(CRef) == tdsa_adamw_chunkref_size());
+  const int chunk_elems = 1 << 16;
+  std::vector<Desc> descs(n);
+  std::vector<CRef> chunks;
+  chunks.reserve(1024);
+  for (size_t i = 0; i < n; ++i) {
+    auto& p = params[i];
+    auto& g = grads[i];
+    Desc d;
+    d.p = p.data_ptr();
+    d.g = g.data_ptr();
+    d.m = ms[i].data_ptr<float>();
+    d.v = vs[i].data_ptr<float>();
+    d.master = masters[i].has_value() ? masters[i]->data_ptr<float>() : nullptr;
+    d.numel = p.numel();
+    d.param_bf16 = dtype_flag(p);
+    d.grad_bf16 = dtype_flag(g);
+    descs[i] = d;
+    const long long nch = (d.numel + chunk_elems - 1) / chunk_elems;
+    for (long long c = 0; c < nch; ++c)
+      chunks.push_back({(int)i, (int)c});
+  }
+  const long long desc_bytes = (long long)(n * sizeof(Desc));
+  const long long total = desc_bytes + (long long)(chunks.size() * sizeof(CRef));
+  auto host = at::empty({total}, at::TensorOptions().dtype(at::kByte));
+  std::memcpy(host.data_ptr(), descs.data(), desc_bytes);
+  std::memcpy((char*)host.data_ptr() + desc_bytes, chunks.data(),
+              chunks.size() * sizeof(CRef));
+  auto dev = host.to(params[0].device());
+  check_hip(tdsa_adamw_multi(dev.data_ptr(), desc_bytes, (int)chunks.size(),
+                             chunk_elems, (float)lr, (float)b1, (float)b2,
+                             (float)eps, (float)wd, step, cur_stream()),
+            "adamw_step_multi");
+}
+
 void sgd_step(at::Tensor param, at::Tensor grad, at::Tensor buf,
               at::Tensor master, bool has_buf, bool has_master, double lr,
               double momentum, double dampening, double wd, bool nesterov,
@@ -357,6 +416,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("cross_entropy_fwd", &cross_entropy_fwd);
   mod.def("cross_entropy_bwd", &cross_entropy_bwd);
   mod.def("adamw_step", &adamw_step);
+  mod.def("adamw_step_multi", &adamw_step_multi);
   mod.def("sgd_step", &sgd_step);
   mod.def("attention_fwd", &attention_fwd, py::arg("q"), py::arg("k"),
           py::arg("v"), py::arg("scale"), py::arg("out") = py::none());

@@ -67,6 +67,28 @@ DEV_INLINE float block_sum(float v, float* scratch) {
   __syncthreads();
   return r;
 }
+// Paired reduction: one barrier round for two sums (layernorm's sum/sumsq
+// and c1/c2 pairs — halves the barrier count of two block_sum calls).
+DEV_INLINE void block_sum2(float& a, float& b, float* scratch) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int nw = blockDim.x / WAVE;
+  a = wave_sum(a);
+  b = wave_sum(b);
+  if (lane == 0) {
+    scratch[wid] = a;
+    scratch[nw + wid] = b;
+  }
+  __syncthreads();
+  float ra = (lane < nw) ? scratch[lane] : 0.0f;
+  float rb = (lane < nw) ? scratch[nw + lane] : 0.0f;
+  ra = wave_sum(ra);
+  rb = wave_sum(rb);
+  a = __shfl(ra, 0, WAVE);
+  b = __shfl(rb, 0, WAVE);
+  __syncthreads();
+}
+
 DEV_INLINE float block_max(float v, float* scratch) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;

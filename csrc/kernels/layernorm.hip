@@ -36,7 +36,7 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
                               int M, int N, float eps) {
   using VT = VecTraits<T>;
   constexpr int W = VT::W;
-  __shared__ float scratch[1024 / WAVE];
+  __shared__ float scratch[2 * 1024 / WAVE];
   const int tid = threadIdx.x;
   const int nth = blockDim.x;
   for (int row = blockIdx.x; row < M; row += gridDim.x) {
@@ -51,8 +51,7 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
         ss += f * f;
       }
     }
-    s = block_sum(s, scratch);
-    ss = block_sum(ss, scratch);
+    block_sum2(s, ss, scratch);
     const float mu = s / N;
     const float var = fmaxf(ss / N - mu * mu, 0.0f);
     const float rs = rsqrtf(var + eps);
@@ -87,7 +86,7 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__
                                  int M, int N) {
   using VT = VecTraits<T>;
   constexpr int W = VT::W;
-  __shared__ float scratch[1024 / WAVE];
+  __shared__ float scratch[2 * 1024 / WAVE];
   const int tid = threadIdx.x;
   const int nth = blockDim.x;
   float accdw[MAXITER][W];
@@ -121,8 +120,9 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__
         }
       }
     }
-    c1 = block_sum(c1, scratch) / N;
-    c2 = block_sum(c2, scratch) / N;
+    block_sum2(c1, c2, scratch);
+    c1 /= N;
+    c2 /= N;
     T* dxr = dx + (long long)row * N;
     for (int i = tid * W; i < N; i += nth * W) {
       typename VT::V dv = VT::load(dyr + i);
@@ -184,7 +184,7 @@ hipError_t tdsa_ln_fwd(const void* x, const void* w, const void* b, void* y,
                        float* mean, float* rstd, int M, int N, float eps,
                        int is_bf16, hipStream_t stream) {
   const int block = 256;
-  const int grid = (M < 8 * 256) ? M : 8 * 256;
+  const int grid = (M < 4096) ? M : 4096;
   if (is_bf16) {
     if (N % 8) return hipErrorInvalidValue;
     hipLaunchKernelGGL(ln_fwd_kernel<bf16>, dim3(grid), dim3(block), 0, stream,
@@ -202,7 +202,7 @@ hipError_t tdsa_ln_fwd(const void* x, const void* w, const void* b, void* y,
 // G (stripe count) is chosen here and reported to the caller so it can size
 // pdw/pdb; call with pdw==nullptr to query G.
 int tdsa_ln_bwd_dx_stripes(int M) {
-  int g = M < 1024 ? M : 1024;
+  int g = M < 2048 ? M : 2048;
   return g < 1 ? 1 : g;
 }
 

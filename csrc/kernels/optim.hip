@@ -61,9 +61,92 @@ __global__ void sgd_kernel(PT* __restrict__ param, const GT* __restrict__ grad,
   }
 }
 
+// ---- multi-tensor AdamW ---------------------------------------------------
+// One launch updates every parameter: a per-parameter python loop costs
+// ~200 launches x 15us per step (measured). Descriptors + chunk table are
+// built host-side and shipped as one small device blob.
+struct AdamTensorDesc {
+  void* p;
+  const void* g;
+  float* m;
+  float* v;
+  float* master;     // nullptr when param is fp32
+  long long numel;
+  int param_bf16;
+  int grad_bf16;
+};
+
+struct ChunkRef {
+  int tensor;
+  int chunk;
+};
+
+template <typename PT, typename GT>
+DEV_INLINE void adamw_update_span(PT* p, const GT* g, float* m, float* v,
+                                  float* master, long long start, long long end,
+                                  float lr, float b1, float b2, float eps,
+                                  float wd, float inv_bc1, float inv_bc2) {
+  for (long long i = start + threadIdx.x; i < end; i += blockDim.x) {
+    float gv = (float)g[i];
+    float pv = master ? master[i] : (float)p[i];
+    pv *= (1.0f - lr * wd);
+    float mi = m[i] = m[i] * b1 + (1.0f - b1) * gv;
+    float vi = v[i] = v[i] * b2 + (1.0f - b2) * gv * gv;
+    pv -= lr * inv_bc1 * mi / (sqrtf(vi * inv_bc2) + eps);
+    if (master) master[i] = pv;
+    p[i] = (PT)pv;
+  }
+}
+
+__global__ void adamw_multi_kernel(const AdamTensorDesc* __restrict__ descs,
+                                   const ChunkRef* __restrict__ chunks,
+                                   int nchunks, int chunk_elems, float lr,
+                                   float b1, float b2, float eps, float wd,
+                                   float inv_bc1, float inv_bc2) {
+  for (int cid = blockIdx.x; cid < nchunks; cid += gridDim.x) {
+    const ChunkRef ch = chunks[cid];
+    const AdamTensorDesc d = descs[ch.tensor];
+    const long long start = (long long)ch.chunk * chunk_elems;
+    const long long end = min(d.numel, start + chunk_elems);
+    if (d.param_bf16 && d.grad_bf16)
+      adamw_update_span((bf16*)d.p, (const bf16*)d.g, d.m, d.v, d.master,
+                        start, end, lr, b1, b2, eps, wd, inv_bc1, inv_bc2);
+    else if (d.param_bf16)
+      adamw_update_span((bf16*)d.p, (const float*)d.g, d.m, d.v, d.master,
+                        start, end, lr, b1, b2, eps, wd, inv_bc1, inv_bc2);
+    else if (d.grad_bf16)
+      adamw_update_span((float*)d.p, (const bf16*)d.g, d.m, d.v, d.master,
+                        start, end, lr, b1, b2, eps, wd, inv_bc1, inv_bc2);
+    else
+      adamw_update_span((float*)d.p, (const float*)d.g, d.m, d.v, d.master,
+                        start, end, lr, b1, b2, eps, wd, inv_bc1, inv_bc2);
+  }
+}
+
 }  // namespace
 
 extern "C" {
+
+// blob layout: [ndescs x AdamTensorDesc][nchunks x ChunkRef], device memory.
+hipError_t tdsa_adamw_multi(const void* blob, long long desc_bytes, int nchunks,
+                            int chunk_elems, float lr, float b1, float b2,
+                            float eps, float wd, long long step,
+                            hipStream_t stream) {
+  const AdamTensorDesc* descs = (const AdamTensorDesc*)blob;
+  const ChunkRef* chunks =
+      (const ChunkRef*)((const char*)blob + desc_bytes);
+  const float inv_bc1 = 1.0f / (1.0f - powf(b1, (float)step));
+  const float inv_bc2 = 1.0f / (1.0f - powf(b2, (float)step));
+  int grid = nchunks < 2048 ? nchunks : 2048;
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(adamw_multi_kernel, dim3(grid), dim3(256), 0, stream,
+                     descs, chunks, nchunks, chunk_elems, lr, b1, b2, eps, wd,
+                     inv_bc1, inv_bc2);
+  return hipGetLastError();
+}
+
+int tdsa_adamw_desc_size() { return (int)sizeof(AdamTensorDesc); }
+int tdsa_adamw_chunkref_size() { return (int)sizeof(ChunkRef); }
 
 hipError_t tdsa_adamw_step(void* param, const void* grad, float* m, float* v,
                            float* master, float* vmax, int has_master,

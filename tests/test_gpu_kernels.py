@@ -264,3 +264,32 @@ def test_fused_qkv_attention_gpu():
     _close(y, ref, 2e-2, "fused attn fwd")
     ref.backward(dy.float())
     _close(qkv.grad, qkv_f.grad, 4e-2, "fused attn bwd")
+
+
+def test_adamw_multi_matches_single_gpu():
+    from tiny_deepspeed_amd import AdamW
+
+    torch.manual_seed(0)
+    shapes = [(128, 64), (3,), (1000,), (64, 64, 2)]
+    p1 = [torch.nn.Parameter(torch.randn(s, device="cuda", dtype=torch.bfloat16))
+          for s in shapes]
+    p2 = [torch.nn.Parameter(t.detach().clone()) for t in p1]
+    o1 = AdamW([(f"p{i}", p) for i, p in enumerate(p1)], lr=1e-2)
+    o2 = AdamW([(f"p{i}", p) for i, p in enumerate(p2)], lr=1e-2)
+    for it in range(3):
+        for a, b in zip(p1, p2):
+            g = torch.randn_like(a)
+            a.grad = g
+            b.grad = g.clone()
+        o1.step()                       # multi-tensor fused path
+        o2._apply_updates = lambda items: [o2.one_step(n, p) for n, p in items]
+        o2.t += 1
+        o2.pre_step()
+        o2._apply_updates([(n, p) for n, p in o2.params.items()
+                           if p.grad is not None])
+        for p in o2.params.values():
+            p.grad = None
+    for a, b in zip(p1, p2):
+        _close(a.data, b.data, 1e-6, "adamw multi vs single")
+    for n in o1.master:
+        _close(o1.master[n], o2.master[n], 1e-6, "master multi vs single")

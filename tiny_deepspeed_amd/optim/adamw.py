@@ -47,6 +47,23 @@ class AdamW(Optimizer):
             self.master[name] = p.detach().float().clone()
 
     @torch.no_grad()
+    def _apply_updates(self, items):
+        if (not items or self.amsgrad or not items[0][1].is_cuda
+                or not ops.ext_available()):
+            return super()._apply_updates(items)
+        params, grads, ms, vs, masters = [], [], [], [], []
+        for name, p in items:
+            params.append(p.data)
+            grads.append(p.grad)
+            ms.append(self.exp_avg[name])
+            vs.append(self.exp_avg_sq[name])
+            masters.append(self.master.get(name))
+        ops.get_ext().adamw_step_multi(
+            params, grads, ms, vs, masters, self.lr, self.beta1, self.beta2,
+            self.eps, self.weight_decay, self.t,
+        )
+
+    @torch.no_grad()
     def one_step(self, name, param):
         ops.adamw_step(
             param.data, param.grad, self.exp_avg[name], self.exp_avg_sq[name],

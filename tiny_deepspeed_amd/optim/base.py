@@ -40,12 +40,19 @@ class Optimizer:
     def step(self):
         self.t += 1
         self.pre_step()
-        for name, param in self.params.items():
-            if self._should_update(name, param):
-                self.one_step(name, param)
+        self._apply_updates(
+            [(n, p) for n, p in self.params.items()
+             if self._should_update(n, p)]
+        )
         self.post_step()
         for param in self.params.values():
             param.grad = None
+
+    def _apply_updates(self, items):
+        """Default: per-parameter updates. AdamW overrides this with one
+        fused multi-tensor kernel launch on GPU (csrc/kernels/optim.hip)."""
+        for name, param in items:
+            self.one_step(name, param)
 
     @torch.no_grad()
     def zero_grad(self):
