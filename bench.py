@@ -21,6 +21,18 @@ from collections import OrderedDict
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
+# hipBLASLt/rocBLAS algorithm selection: use the committed TunableOp cache
+# (read-only — tuning itself takes minutes and was done offline; +4% over
+# the default heuristics on the GPT-2 medium shapes). torch inserts the
+# device ordinal before the extension, so tuned/ ships one copy per GPU.
+_TUNED = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                      "tuned", "tunableop.csv")
+if (os.path.exists(_TUNED.replace(".csv", "0.csv"))
+        and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ):
+    os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
+    os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
+    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = _TUNED
+
 import torch
 import torch.distributed as dist
 
@@ -164,6 +176,9 @@ def main():
                 "global_batch": args.batch * world_size,
                 "seq_len": args.seq,
                 "parallelism": f"{args.parallel}-dp{world_size}",
+                "peak_hbm_gb": round(
+                    torch.cuda.max_memory_allocated() / 2**30, 2
+                ) if have_gpu else None,
             },
         }
         print(json.dumps(out))

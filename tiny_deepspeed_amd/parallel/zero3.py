@@ -35,8 +35,19 @@ from .wrapper import ModelWrapper
 
 
 class _GatherMixin:
-    def _gather(self, param):
-        """Return the full tensor for `param` (broadcast from owner)."""
+    """JIT parameter gathers with one-module-ahead prefetch.
+
+    The naive protocol (issue broadcast, wait, compute) leaves the xGMI
+    transfer on the critical path of every layer. Here each callback
+    (a) consumes buffers whose broadcasts were issued earlier, (b) waits the
+    gather stream — at that point ONLY this module's gathers are
+    outstanding — then (c) issues the next module's gathers, which ride the
+    gather stream/communicator underneath this module's compute. The
+    wrapper links modules in traversal order (== execution order for
+    transformer stacks); backward prefetches in reverse.
+    """
+
+    def _gather_async(self, param):
         if param is None:
             return None
         comm = self._comm
@@ -50,18 +61,39 @@ class _GatherMixin:
         comm.gather_broadcast(buf, src=param._tdsa_owner)
         return buf
 
+    def _issue_gathers(self):
+        # epoch guard: buffers prefetched during backward must not survive
+        # the optimizer step into the next forward (stale parameters)
+        epoch = getattr(self._comm, "gather_epoch", 0)
+        if (getattr(self, "_tdsa_bufs", None) is None
+                or getattr(self, "_tdsa_buf_epoch", -1) != epoch):
+            object.__setattr__(self, "_tdsa_bufs", {
+                pname: self._gather_async(getattr(self, pname, None))
+                for pname in ("weight", "bias")
+            })
+            object.__setattr__(self, "_tdsa_buf_epoch", epoch)
+
+    def _take_bufs(self, direction):
+        """Wait for this module's gathered params; prefetch the neighbor's."""
+        self._issue_gathers()
+        self._comm.wait_gather()
+        bufs = self._tdsa_bufs
+        object.__setattr__(self, "_tdsa_bufs", None)
+        nxt = getattr(self, direction, None)
+        if nxt is not None:
+            nxt._issue_gathers()
+        return bufs
+
 
 class Linear(_GatherMixin, base.Linear):
     def forward_callback(self, x, weight, bias):
-        w = self._gather(self.weight)
-        b = self._gather(self.bias)
-        self._comm.wait_gather()
-        return ops.linear_forward(x, w, b, tuner=self.tuner)
+        bufs = self._take_bufs("_tdsa_next")
+        return ops.linear_forward(x, bufs["weight"], bufs["bias"],
+                                  tuner=self.tuner)
         # gather buffers die here; the allocator reclaims them stream-safely
 
     def backward_callback(self, dy, x):
-        w = self._gather(self.weight)
-        self._comm.wait_gather()
+        w = self._take_bufs("_tdsa_prev")["weight"]
         if self.weight.requires_grad:
             dw = ops.linear_weight_grad(dy, x, tuner=self.tuner)
             assert tuple(dw.shape) == self.weight._tdsa_full_shape
@@ -75,14 +107,12 @@ class Linear(_GatherMixin, base.Linear):
 
 class LayerNorm(_GatherMixin, base.LayerNorm):
     def forward_callback(self, x, weight, bias):
-        w = self._gather(self.weight)
-        b = self._gather(self.bias)
-        self._comm.wait_gather()
-        return ops.layernorm_fwd(x, w, b, eps=self.eps, tuner=self.tuner)
+        bufs = self._take_bufs("_tdsa_next")
+        return ops.layernorm_fwd(x, bufs["weight"], bufs["bias"],
+                                 eps=self.eps, tuner=self.tuner)
 
     def backward_callback(self, dy, x, mean, rstd):
-        w = self._gather(self.weight)
-        self._comm.wait_gather()
+        w = self._take_bufs("_tdsa_prev")["weight"]
         dx, ws = ops.layernorm_dx(dy, x, w, mean, rstd, tuner=self.tuner)
         if self.weight.requires_grad:
             dw, db = ops.layernorm_dwdb(ws, dtype=self.weight.dtype,
@@ -94,8 +124,7 @@ class LayerNorm(_GatherMixin, base.LayerNorm):
 
 class Embedding(_GatherMixin, base.Embedding):
     def forward_callback(self, idx, weight):
-        w = self._gather(self.weight)
-        self._comm.wait_gather()
+        w = self._take_bufs("_tdsa_next")["weight"]
         return ops.embedding_forward(w, idx, padding_idx=self.padding_idx,
                                      tuner=self.tuner)
 
@@ -152,7 +181,19 @@ class Zero3(ModelWrapper):
         self._shard_device = device
         super().__init__(module, parts=parts, comm=comm)
 
+    def forward(self, *args, **kwargs):
+        self.comm.gather_epoch = getattr(self.comm, "gather_epoch", 0) + 1
+        return super().forward(*args, **kwargs)
+
     def _post_wrap(self):
+        # link wrapped modules in traversal order for gather prefetch
+        chain = [m for m in self.module.modules()
+                 if isinstance(m, _GatherMixin)]
+        for a, b in zip(chain, chain[1:]):
+            # bypass nn.Module.__setattr__: these links are scheduling
+            # hints, not submodules
+            object.__setattr__(a, "_tdsa_next", b)
+            object.__setattr__(b, "_tdsa_prev", a)
         rank = self.comm.rank
         if self._shard_device is not None:
             dev = torch.device(self._shard_device)
