@@ -53,6 +53,7 @@ hipError_t tdsa_attn_bwd(const void*, const void*, const void*, const void*,
                          const long long*, const long long*, hipStream_t);
 hipError_t tdsa_dbg_mfma(const void*, const void*, float*, int, hipStream_t);
 hipError_t tdsa_dbg_stage(const void*, void*, int, hipStream_t);
+hipError_t tdsa_dbg_tr16(const void*, float*, hipStream_t);
 }
 
 namespace {
@@ -400,11 +401,20 @@ at::Tensor dbg_stage(at::Tensor in, int64_t transposed) {
   return out;
 }
 
+at::Tensor dbg_tr16(at::Tensor in) {
+  CHECK_IN(in);
+  auto out = at::zeros({2, 4, 2, 4, 64}, in.options().dtype(at::kFloat));
+  check_hip(tdsa_dbg_tr16(in.data_ptr(), out.data_ptr<float>(), cur_stream()),
+            "dbg_tr16");
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("dbg_mfma", &dbg_mfma);
   mod.def("dbg_stage", &dbg_stage);
+  mod.def("dbg_tr16", &dbg_tr16);
   mod.def("layernorm_fwd", &layernorm_fwd);
   mod.def("layernorm_bwd_dx", &layernorm_bwd_dx);
   mod.def("layernorm_bwd_dwdb", &layernorm_bwd_dwdb);

@@ -102,39 +102,6 @@ DEV_INLINE void stage_rowmajor(const bf16* __restrict__ g, char* lds,
         load8(g + a.stage_src[rep]);
 }
 
-// Transposed staging (dest row = source col), scalar u16 writes.
-template <int NT>
-DEV_INLINE void stage_transposed(const bf16* __restrict__ g, char* lds,
-                                 const TileAddr& a, const int (&tdst)[4][8]) {
-#pragma unroll
-  for (int rep = 0; rep < 512 / NT; ++rep) {
-    short8v v = load8(g + a.stage_src[rep]);
-#pragma unroll
-    for (int j = 0; j < 8; ++j)
-      *reinterpret_cast<short*>(lds + tdst[rep][j]) = v[j];
-  }
-}
-
-// Transposed images use a second XOR term (row bits 3-5): during the scalar
-// write burst the dest row is congruent mod 8 across a half-wave, so the
-// plain swizzle degenerates to f(source-row-only) banks (16-way measured,
-// 13.5% of dkv wave cycles); row bits 3-5 vary across the burst and spread
-// it to <=2-way. Readers XOR the same term (swzT(r,b) = swz(r,b) ^ trx(r)).
-DEV_INLINE int trx(int row) { return ((row >> 3) & 7) << 4; }
-
-template <int NT>
-DEV_INLINE void transposed_dst(int tid, int (&tdst)[4][8]) {
-#pragma unroll
-  for (int rep = 0; rep < 512 / NT; ++rep) {
-    int chunk = tid + rep * NT;
-    int row = chunk >> 3;
-    int c0 = (chunk & 7) * 8;
-#pragma unroll
-    for (int j = 0; j < 8; ++j)
-      tdst[rep][j] = swz(c0 + j, row * 2) ^ trx(c0 + j);
-  }
-}
-
 // C-layout -> A-fragment repack. Values live per lane as 16 C registers per
 // 32-wide tile (packed to bf16 word pairs wA[r1]=(r0=0,1), wB[r1]=(r0=2,3)).
 // A-frag slice s (k = 16s + 8*h32 + e at this lane's own 32-axis index)
@@ -159,6 +126,43 @@ DEV_INLINE bfrag frag_from_packed(const PackedC& p) {
   r.w[2] = rA[1];
   r.w[3] = rB[1];
   return r.f;
+}
+
+typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4v;
+#define LDS_AS __attribute__((address_space(3)))
+
+// B-fragment via gfx950 hardware transpose read (ds_read_tr16_b64): two
+// reads deliver k = 16*S + 8*(l/32) + e over ROWS of a row-major swizzled
+// [64][64] image at this lane's column — no transposed LDS image needed
+// (the scalar transposed staging and its bank conflicts are gone).
+// Lane mapping hardware-verified by scripts/debug_mfma.py::probe_tr16.
+DEV_INLINE bfrag tr_bfrag(const char* lds, int a_lo, int a_hi) {
+  bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (LDS_AS bf16x4v*)(lds + a_lo));
+  bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (LDS_AS bf16x4v*)(lds + a_hi));
+  union { bfrag f; bf16x4v h[2]; } r;
+  r.h[0] = lo;
+  r.h[1] = hi;
+  return r.f;
+}
+
+// Per-lane tr-read base addresses: trb[dt][half] for the 32-col tile dt;
+// slice S adds S*2048 bytes (16 rows). Lane i=l&15 supplies the chunk at
+// row (i>>2) of the 4-row block, byte quarter 4*(i&3).
+DEV_INLINE void tr_bases(int lane, int (&trb)[2][2]) {
+  const int g4 = lane >> 4;
+  const int i = lane & 15;
+  const int hh = g4 >> 1;
+  const int iq = i >> 2;
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int row = 8 * hh + 4 * half + iq;
+      const int colbyte = (dt * 32 + 16 * (g4 & 1) + 4 * (i & 3)) * 2;
+      trb[dt][half] = row * 128 + (colbyte ^ ((row & 7) << 4));
+    }
 }
 
 DEV_INLINE bfrag load_frag_scaled(const bf16* p, int row, int col, int st,
@@ -193,7 +197,7 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
   constexpr int NT = NW * WAVE;
   __shared__ __attribute__((aligned(16))) char smem[2 * KVB * D * 2];
   char* lds_k = smem;                 // [64][64] keys row-major
-  char* lds_vt = smem + KVB * D * 2;  // [64(d)][64(key)] V transposed
+  char* lds_v = smem + KVB * D * 2;   // [64][64] values row-major
 
   const int qb = blockIdx.x;
   const long long bh = blockIdx.y;
@@ -209,17 +213,14 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
   const bf16* vp = v + boff;
 
   const TileAddr ta = tile_addr<NT>(tid, sq.t);
-  int tdst[4][8];
-  transposed_dst<NT>(tid, tdst);
-  // fragment offsets; the V^T image shares the K image's pattern
   int kf_off[2][4];  // [32-tile][k-slice]
 #pragma unroll
   for (int t2 = 0; t2 < 2; ++t2)
 #pragma unroll
     for (int s = 0; s < 4; ++s)
       kf_off[t2][s] = swz(t2 * 32 + q32, (s * 16 + 8 * h32) * 2);
-
-  const int xtr[2] = {trx(q32), trx(32 + q32)};
+  int trb[2][2];
+  tr_bases(lane, trb);
   const float qscale = scale * LOG2E;
   bfrag q_frag[4];
 #pragma unroll
@@ -237,7 +238,7 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
   for (int j = 0; j < n_kv; ++j) {
     __syncthreads();
     stage_rowmajor<NT>(kp + (long long)(j * KVB) * sq.t, lds_k, ta);
-    stage_transposed<NT>(vp + (long long)(j * KVB) * sq.t, lds_vt, ta, tdst);
+    stage_rowmajor<NT>(vp + (long long)(j * KVB) * sq.t, lds_v, ta);
     __syncthreads();
 
     const int key0 = j * KVB;
@@ -316,7 +317,8 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
       }
 #pragma unroll
       for (int dt = 0; dt < 2; ++dt)
-        o_acc[dt] = MFMA32(pa, lds_read16(lds_vt, kf_off[dt][s] ^ xtr[dt]),
+        o_acc[dt] = MFMA32(pa, tr_bfrag(lds_v, trb[dt][0] + s * 2048,
+                                        trb[dt][1] + s * 2048),
                            o_acc[dt]);
     }
   }
@@ -374,12 +376,10 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
                                     GStride so, GStride sd) {
   constexpr int BK = NW * 32;
   constexpr int NT = NW * WAVE;
-  __shared__ __attribute__((aligned(16))) char smem[4 * KVB * D * 2 + 2 * KVB * 4];
+  __shared__ __attribute__((aligned(16))) char smem[2 * KVB * D * 2 + 2 * KVB * 4];
   char* lds_q = smem;
   char* lds_do = smem + KVB * D * 2;
-  char* lds_qt = smem + 2 * KVB * D * 2;
-  char* lds_dot = smem + 3 * KVB * D * 2;
-  float* lds_lse = reinterpret_cast<float*>(smem + 4 * KVB * D * 2);
+  float* lds_lse = reinterpret_cast<float*>(smem + 2 * KVB * D * 2);
   float* lds_dlt = lds_lse + KVB;
 
   const int jb = blockIdx.x;
@@ -399,16 +399,14 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
 
   const TileAddr ta = tile_addr<NT>(tid, sq.t);
   const TileAddr tao = tile_addr<NT>(tid, so.t);
-  int tdst[4][8];
-  transposed_dst<NT>(tid, tdst);
-  int af_off[2][4];  // shared by all four LDS images (same pattern)
+  int af_off[2][4];
 #pragma unroll
   for (int t2 = 0; t2 < 2; ++t2)
 #pragma unroll
     for (int s = 0; s < 4; ++s)
       af_off[t2][s] = swz(t2 * 32 + k32, (s * 16 + 8 * h32) * 2);
-
-  const int xtr[2] = {trx(k32), trx(32 + k32)};
+  int trb[2][2];
+  tr_bases(lane, trb);
   const float kscale = scale * LOG2E;
   const int key_lo = jb * BK + w * 32;
   const int key_me = key_lo + k32;
@@ -427,8 +425,6 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
     __syncthreads();
     stage_rowmajor<NT>(qp + (long long)(i * KVB) * sq.t, lds_q, ta);
     stage_rowmajor<NT>(dop + (long long)(i * KVB) * so.t, lds_do, tao);
-    stage_transposed<NT>(qp + (long long)(i * KVB) * sq.t, lds_qt, ta, tdst);
-    stage_transposed<NT>(dop + (long long)(i * KVB) * so.t, lds_dot, tao, tdst);
     if (tid < KVB) {
       lds_lse[tid] = lse[bh * T + i * KVB + tid] * LOG2E;
       lds_dlt[tid] = delta[bh * T + i * KVB + tid];
@@ -482,9 +478,11 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
       }
 #pragma unroll
       for (int dt = 0; dt < 2; ++dt) {
-        dv_acc[dt] = MFMA32(ap, lds_read16(lds_dot, af_off[dt][s] ^ xtr[dt]),
+        dv_acc[dt] = MFMA32(ap, tr_bfrag(lds_do, trb[dt][0] + s * 2048,
+                                         trb[dt][1] + s * 2048),
                             dv_acc[dt]);
-        dk_acc[dt] = MFMA32(as, lds_read16(lds_qt, af_off[dt][s] ^ xtr[dt]),
+        dk_acc[dt] = MFMA32(as, tr_bfrag(lds_q, trb[dt][0] + s * 2048,
+                                         trb[dt][1] + s * 2048),
                             dk_acc[dt]);
       }
     }
@@ -519,10 +517,9 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
                                    GStride sd) {
   constexpr int BM = NW * 32;
   constexpr int NT = NW * WAVE;
-  __shared__ __attribute__((aligned(16))) char smem[3 * KVB * D * 2];
-  char* lds_k = smem;                     // K row-major (A of S^T)
-  char* lds_kt = smem + KVB * D * 2;      // K^T (B of dQ)
-  char* lds_v = smem + 2 * KVB * D * 2;   // V row-major (A of dP^T)
+  __shared__ __attribute__((aligned(16))) char smem[2 * KVB * D * 2];
+  char* lds_k = smem;                     // K row-major
+  char* lds_v = smem + KVB * D * 2;       // V row-major
 
   const int qb = blockIdx.x;
   const long long bh = blockIdx.y;
@@ -540,16 +537,14 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   const bf16* dop = dout + ooff + (long long)(qb * BM) * so.t;
 
   const TileAddr ta = tile_addr<NT>(tid, sq.t);
-  int tdst[4][8];
-  transposed_dst<NT>(tid, tdst);
   int f_off[2][4];
 #pragma unroll
   for (int t2 = 0; t2 < 2; ++t2)
 #pragma unroll
     for (int s = 0; s < 4; ++s)
       f_off[t2][s] = swz(t2 * 32 + q32, (s * 16 + 8 * h32) * 2);
-
-  const int xtr[2] = {trx(q32), trx(32 + q32)};
+  int trb[2][2];
+  tr_bases(lane, trb);
   const float qscale = scale * LOG2E;
   bfrag q_frag[4], do_frag[4];
 #pragma unroll
@@ -569,7 +564,6 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   for (int j = 0; j < n_kv; ++j) {
     __syncthreads();
     stage_rowmajor<NT>(kp + (long long)(j * KVB) * sq.t, lds_k, ta);
-    stage_transposed<NT>(kp + (long long)(j * KVB) * sq.t, lds_kt, ta, tdst);
     stage_rowmajor<NT>(vp + (long long)(j * KVB) * sq.t, lds_v, ta);
     __syncthreads();
 
@@ -615,7 +609,8 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
       }
 #pragma unroll
       for (int dt = 0; dt < 2; ++dt)
-        dq_acc[dt] = MFMA32(as, lds_read16(lds_kt, f_off[dt][s] ^ xtr[dt]),
+        dq_acc[dt] = MFMA32(as, tr_bfrag(lds_k, trb[dt][0] + s * 2048,
+                                         trb[dt][1] + s * 2048),
                             dq_acc[dt]);
     }
   }

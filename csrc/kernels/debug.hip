@@ -68,9 +68,51 @@ __global__ void dbg_stage_kernel(const bf16* __restrict__ in,
   }
 }
 
+typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4v;
+#define LDS_AS __attribute__((address_space(3)))
+
+// Probe ds_read_tr16_b64 lane mapping: stage a [64][64] bf16 image (swizzled
+// row-major), then issue tr reads with the attention kernels' intended
+// addressing and dump every (dt, s, half, j, lane) result.
+__global__ void dbg_tr16_kernel(const bf16* __restrict__ in,
+                                float* __restrict__ out) {
+  __shared__ __attribute__((aligned(16))) char lds[64 * 64 * 2];
+#pragma unroll
+  for (int rep = 0; rep < 8; ++rep) {  // 64 threads stage 512 chunks
+    int chunk = threadIdx.x + rep * 64;
+    int row = chunk >> 3;
+    int c0 = (chunk & 7) * 8;
+    *reinterpret_cast<short8v*>(lds + swz(row, c0 * 2)) = load8(in + chunk * 8);
+  }
+  __syncthreads();
+  const int l = threadIdx.x;
+  const int g4 = l >> 4;
+  const int i = l & 15;
+  for (int dt = 0; dt < 2; ++dt)
+    for (int s = 0; s < 4; ++s)
+      for (int half = 0; half < 2; ++half) {
+        // intended block: rows 16s + 8*(g4>>1) + 4*half + (i/4),
+        //                 cols dt*32 + 16*(g4&1) + 4*(i%4)
+        const int row = 16 * s + 8 * (g4 >> 1) + 4 * half + (i >> 2);
+        const int colbyte = (dt * 32 + 16 * (g4 & 1) + 4 * (i & 3)) * 2;
+        const int addr = swz(row, colbyte);
+        auto p = (LDS_AS bf16x4v*)(lds + addr);
+        bf16x4v r = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          out[(((dt * 4 + s) * 2 + half) * 4 + j) * 64 + l] = (float)r[j];
+      }
+}
+
 }  // namespace
 
 extern "C" {
+
+hipError_t tdsa_dbg_tr16(const void* in, float* out, hipStream_t stream) {
+  hipLaunchKernelGGL(dbg_tr16_kernel, dim3(1), dim3(64), 0, stream,
+                     (const bf16*)in, out);
+  return hipGetLastError();
+}
 
 hipError_t tdsa_dbg_mfma(const void* A, const void* B, float* D, int variant,
                          hipStream_t stream) {

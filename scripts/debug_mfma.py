@@ -43,5 +43,35 @@ def main():
     print(f"stage transposed err vs M^T: {err_t:.4f} (expected 0)")
 
 
+def probe_tr16():
+    import torch
+    from tiny_deepspeed_amd import _C
+    dev = "cuda"
+    Mr = torch.arange(64, device=dev).view(64, 1).expand(64, 64).contiguous().to(torch.bfloat16)
+    Mc = torch.arange(64, device=dev).view(1, 64).expand(64, 64).contiguous().to(torch.bfloat16)
+    Rr = _C.dbg_tr16(Mr)  # [dt][s][half][j][lane] -> row index read
+    Rc = _C.dbg_tr16(Mc)  # -> col index read
+    # hypothesis: lane l gets row 16s+8*(g4>>1)+4*half+j, col dt*32+16*(g4&1)+(l&15)
+    import itertools
+    ok = True
+    for dt, s, half, j in itertools.product(range(2), range(4), range(2), range(4)):
+        for l in range(64):
+            g4 = l >> 4
+            er = 16*s + 8*(g4 >> 1) + 4*half + j
+            ec = dt*32 + 16*(g4 & 1) + (l & 15)
+            ar = int(Rr[dt, s, half, j, l].item())
+            ac = int(Rc[dt, s, half, j, l].item())
+            if (ar, ac) != (er, ec):
+                if ok:
+                    print(f"tr16 MISMATCH first at dt{dt} s{s} h{half} j{j} l{l}: got (r{ar},c{ac}) want (r{er},c{ec})")
+                ok = False
+    print("tr16 hypothesis:", "OK" if ok else "FAILED")
+    if not ok:
+        # dump mapping for (dt,s,half)=(0,0,0)
+        for j in range(4):
+            print("j", j, [(int(Rr[0,0,0,j,l]), int(Rc[0,0,0,j,l])) for l in range(0,64,4)])
+
+
 if __name__ == "__main__":
     main()
+    probe_tr16()
