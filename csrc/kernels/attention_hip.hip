@@ -505,7 +505,7 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
 // Backward dQ: one workgroup per (NW*32)-row Q block; wave w owns 32 rows.
 // ---------------------------------------------------------------------------
 template <int NW>
-__launch_bounds__(NW * WAVE)
+__launch_bounds__(NW * WAVE)  // capping at 128 VGPR spills 164 B/lane here
 __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
                                    const bf16* __restrict__ k,
                                    const bf16* __restrict__ v,
