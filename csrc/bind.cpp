@@ -124,8 +124,8 @@ std::vector<at::Tensor> layernorm_bwd_dwdb(at::Tensor pdw, at::Tensor pdb) {
   CHECK_IN(pdw); CHECK_IN(pdb);
   const int G = pdw.size(0);
   const int N = pdw.size(1);
-  auto dw = at::zeros({N}, pdw.options());
-  auto db = at::zeros({N}, pdb.options());
+  auto dw = at::empty({N}, pdw.options());
+  auto db = at::empty({N}, pdb.options());
   check_hip(tdsa_ln_bwd_dwdb(pdw.data_ptr<float>(), pdb.data_ptr<float>(),
                              dw.data_ptr<float>(), db.data_ptr<float>(), G, N,
                              cur_stream()),

@@ -156,25 +156,29 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__
 }
 
 // Column-reduce the stripe buffers -> dw[N], db[N] (fp32 out; caller casts).
-// 2-D grid: x covers columns, y covers stripe chunks; one fp32 atomicAdd per
-// (chunk, column) into the zero-initialized outputs (Guideline 12: per-block
-// partials first). A 1-D thread-per-column grid is N/256 workgroups — far
-// under the 256 CUs — and was the top kernel in the first bench profile.
+// One wave per column (4 columns per block): lanes stride the G stripes,
+// wave-reduce, lane 0 writes. No atomics, no zero-init launch (an earlier
+// 2-D atomicAdd version needed two at::zeros fills per call — ~100 fill
+// launches per step; and the first 1-D thread-per-column version was the
+// single worst kernel in the bench profile at N/256 workgroups).
 __global__ void ln_bwd_dwdb_kernel(const float* __restrict__ pdw,
                                    const float* __restrict__ pdb,
                                    float* __restrict__ dw, float* __restrict__ db,
-                                   int G, int N, int chunk) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+                                   int G, int N) {
+  const int c = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
   if (c >= N) return;
-  const int g0 = blockIdx.y * chunk;
-  const int g1 = min(G, g0 + chunk);
+  const int lane = threadIdx.x & (WAVE - 1);
   float sw = 0.f, sb = 0.f;
-  for (int g = g0; g < g1; ++g) {
+  for (int g = lane; g < G; g += WAVE) {
     sw += pdw[(long long)g * N + c];
     sb += pdb[(long long)g * N + c];
   }
-  atomicAdd(&dw[c], sw);
-  atomicAdd(&db[c], sb);
+  sw = wave_sum(sw);
+  sb = wave_sum(sb);
+  if (lane == 0) {
+    dw[c] = sw;
+    db[c] = sb;
+  }
 }
 
 }  // namespace
@@ -227,16 +231,13 @@ hipError_t tdsa_ln_bwd_dx(const void* dy, const void* x, const void* w,
   return hipGetLastError();
 }
 
-// dw/db must be zero-initialized by the caller.
 hipError_t tdsa_ln_bwd_dwdb(const float* pdw, const float* pdb, float* dw,
                             float* db, int G, int N, hipStream_t stream) {
   const int block = 256;
-  const int gx = (N + block - 1) / block;
-  int gy = 1;
-  while (gx * gy < 1024 && gy * 8 < G) gy *= 2;
-  const int chunk = (G + gy - 1) / gy;
-  hipLaunchKernelGGL(ln_bwd_dwdb_kernel, dim3(gx, gy), dim3(block), 0, stream,
-                     pdw, pdb, dw, db, G, N, chunk);
+  const int cols_per_block = block / WAVE;
+  const int grid = (N + cols_per_block - 1) / cols_per_block;
+  hipLaunchKernelGGL(ln_bwd_dwdb_kernel, dim3(grid), dim3(block), 0, stream,
+                     pdw, pdb, dw, db, G, N);
   return hipGetLastError();
 }
 

@@ -41,8 +41,9 @@ class _ZeroOptimMixin:
 
     def post_step(self):
         if self.broadcast_params_after_step and self.comm.world_size > 1:
-            # refresh replicas: async broadcast of every param from its
-            # owner, identical order on all ranks, then one stream wait
-            for name, param in self.params.items():
-                self.comm.broadcast(param.data, src=self._owner(name))
+            # refresh replicas: bucketed async broadcasts from each owner,
+            # identical order on all ranks, then one stream wait
+            self.comm.broadcast_bucketed(
+                [(p.data, self._owner(n)) for n, p in self.params.items()]
+            )
             self.comm.sync()

@@ -145,6 +145,69 @@ class CommContext:
     def gather_broadcast(self, t, src):
         return self.broadcast(t, src, channel=GATHER)
 
+    def broadcast_bucketed(self, tensors_with_src, bucket_bytes=1 << 22,
+                           channel=REDUCE):
+        """Broadcast many tensors, coalescing consecutive SMALL same-owner
+        tensors into flat buckets (one collective each). Addresses the
+        reference's 'communication bucketing' TODO (README.md:71): the GPT-2
+        census has ~100 sub-1MB payloads whose per-collective launch latency
+        dominates their transfer time. Large tensors broadcast directly
+        (a flatten round trip would double their local traffic)."""
+        if not self.initialized or self.world_size == 1:
+            return
+        groups = []  # (src, [tensors]) of consecutive small same-src tensors
+        for t, src in tensors_with_src:
+            nbytes = t.numel() * t.element_size()
+            if nbytes >= bucket_bytes:
+                groups.append((src, None, t))
+                continue
+            if (groups and groups[-1][1] is not None
+                    and groups[-1][0] == src
+                    and groups[-1][2] + nbytes <= bucket_bytes
+                    and groups[-1][1][0].dtype == t.dtype):
+                groups[-1][1].append(t)
+                groups[-1] = (src, groups[-1][1], groups[-1][2] + nbytes)
+            else:
+                groups.append((src, [t], nbytes))
+        for src, bucket, t_or_bytes in groups:
+            if bucket is None:
+                self.broadcast(t_or_bytes, src, channel=channel)
+            elif len(bucket) == 1:
+                self.broadcast(bucket[0], src, channel=channel)
+            else:
+                self._broadcast_flat(bucket, src, channel)
+
+    def _broadcast_flat(self, bucket, src, channel):
+        stream = self.streams[channel]
+        numels = [t.numel() for t in bucket]
+
+        def body(async_op=False):
+            if self.rank == src:
+                flat = torch.cat([t.reshape(-1) for t in bucket])
+            else:
+                flat = torch.empty(sum(numels), dtype=bucket[0].dtype,
+                                   device=bucket[0].device)
+            work = dist.broadcast(flat, src=src, group=self.pg[channel],
+                                  async_op=async_op)
+            if async_op and work is not None:
+                work.wait()  # gloo path: complete before scatter-back
+            if self.rank != src:
+                off = 0
+                for t, n in zip(bucket, numels):
+                    t.view(-1).copy_(flat[off:off + n])
+                    off += n
+            return flat
+
+        if stream is not None:
+            stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
+                flat = body()
+                flat.record_stream(stream)
+                for t in bucket:
+                    t.record_stream(stream)
+        else:
+            self._keepalive.append(body(async_op=True))
+
     def all_reduce_scalar_avg(self, t):
         """Synchronous scalar average (loss logging)."""
         if not self.initialized or self.world_size == 1:
