@@ -246,6 +246,7 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
 
     // S^T tiles: C[key = 32*t2 + crow(r,h32)][qrow = q32], log2-scaled
     f32x16 st[2];
+    __builtin_amdgcn_s_setprio(1);  // favor MFMA-issuing waves (T5)
 #pragma unroll
     for (int t2 = 0; t2 < 2; ++t2) {
       f32x16 acc = {};
@@ -254,6 +255,7 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
         acc = MFMA32(lds_read16(lds_k, kf_off[t2][s]), q_frag[s], acc);
       st[t2] = acc;
     }
+    __builtin_amdgcn_s_setprio(0);
 
     const bool diag = key0 + KVB - 1 > row_lo;
     float mt = -INFINITY;
@@ -306,6 +308,7 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
     }
 
     // PV
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int s = 0; s < 4; ++s) {
       bfrag pa;
@@ -321,6 +324,7 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
                                         trb[dt][1] + s * 2048),
                            o_acc[dt]);
     }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   float linv_row[16];
@@ -441,11 +445,13 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
       // S, dP tiles: C[qrow = 32*t2 + crow(r,h32)][key = k32]
       f32x16 s_acc = {};
       f32x16 dp_acc = {};
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int s = 0; s < 4; ++s) {
         s_acc = MFMA32(lds_read16(lds_q, af_off[t2][s]), k_frag[s], s_acc);
         dp_acc = MFMA32(lds_read16(lds_do, af_off[t2][s]), v_frag[s], dp_acc);
       }
+      __builtin_amdgcn_s_setprio(0);
 #pragma unroll
       for (int r1 = 0; r1 < 4; ++r1) {
         float p[4], dsv[4];
@@ -467,6 +473,7 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
     }
 
     // dV[key][d] += P^T dO ; dK[key][d] += dS^T Q  (A-frag k = qrow)
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int s = 0; s < 4; ++s) {
       bfrag ap, as;
@@ -486,6 +493,7 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
                             dk_acc[dt]);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   const long long doff = (bh / H) * sd.b + (bh % H) * sd.h
@@ -577,11 +585,13 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
       // S^T, dP^T tiles: C[key = 32*t2 + crow(r,h32)][qrow = q32]
       f32x16 s_acc = {};
       f32x16 dp_acc = {};
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int s = 0; s < 4; ++s) {
         s_acc = MFMA32(lds_read16(lds_k, f_off[t2][s]), q_frag[s], s_acc);
         dp_acc = MFMA32(lds_read16(lds_v, f_off[t2][s]), do_frag[s], dp_acc);
       }
+      __builtin_amdgcn_s_setprio(0);
       float dsv[16];
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
@@ -598,6 +608,7 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
     }
 
     // dQ[qrow][d] += dS K  (A-frag k = key)
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int s = 0; s < 4; ++s) {
       bfrag as;
@@ -613,6 +624,7 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
                                          trb[dt][1] + s * 2048),
                             dq_acc[dt]);
     }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   bf16* dqp = dq + (bh / H) * sd.b + (bh % H) * sd.h
