@@ -13,12 +13,13 @@
 
 // ---- extern "C" launcher prototypes (csrc/kernels/*.hip) ------------------
 extern "C" {
-hipError_t tdsa_ln_fwd(const void*, const void*, const void*, void*, float*,
-                       float*, int, int, float, int, hipStream_t);
+hipError_t tdsa_ln_fwd(const void*, const void*, void*, const void*,
+                       const void*, void*, float*, float*, int, int, float,
+                       int, hipStream_t);
 int tdsa_ln_bwd_dx_stripes(int M);
-hipError_t tdsa_ln_bwd_dx(const void*, const void*, const void*, const float*,
-                          const float*, void*, float*, float*, int, int, int,
-                          hipStream_t);
+hipError_t tdsa_ln_bwd_dx(const void*, const void*, const void*, const void*,
+                          const float*, const float*, void*, float*, float*,
+                          int, int, int, hipStream_t);
 hipError_t tdsa_ln_bwd_dwdb(const float*, const float*, float*, float*, int,
                             int, hipStream_t);
 hipError_t tdsa_gelu_fwd(const void*, void*, long long, int, hipStream_t);
@@ -79,7 +80,8 @@ hipStream_t cur_stream() {
 
 // ---- layernorm ------------------------------------------------------------
 std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
-                                      double eps) {
+                                      double eps,
+                                      c10::optional<at::Tensor> res) {
   CHECK_IN(x); CHECK_IN(w); CHECK_IN(b);
   const int N = x.size(-1);
   const long long M = x.numel() / N;
@@ -87,19 +89,32 @@ std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
   auto f32 = x.options().dtype(at::kFloat);
   auto mean = at::empty({M}, f32);
   auto rstd = at::empty({M}, f32);
-  check_hip(tdsa_ln_fwd(x.data_ptr(), w.data_ptr(), b.data_ptr(), y.data_ptr(),
-                        mean.data_ptr<float>(), rstd.data_ptr<float>(), (int)M,
-                        N, (float)eps, dtype_flag(x), cur_stream()),
+  at::Tensor h;
+  void* res_p = nullptr;
+  void* h_p = nullptr;
+  if (res.has_value()) {
+    CHECK_IN((*res));
+    h = at::empty_like(x);
+    res_p = res->data_ptr();
+    h_p = h.data_ptr();
+  }
+  check_hip(tdsa_ln_fwd(x.data_ptr(), res_p, h_p, w.data_ptr(), b.data_ptr(),
+                        y.data_ptr(), mean.data_ptr<float>(),
+                        rstd.data_ptr<float>(), (int)M, N, (float)eps,
+                        dtype_flag(x), cur_stream()),
             "layernorm_fwd");
   // mean/rstd viewed to the row shape of x
   auto row_sizes = x.sizes().vec();
   row_sizes.pop_back();
+  if (res.has_value())
+    return {y, mean.view(row_sizes), rstd.view(row_sizes), h};
   return {y, mean.view(row_sizes), rstd.view(row_sizes)};
 }
 
 std::vector<at::Tensor> layernorm_bwd_dx(at::Tensor dy, at::Tensor x,
                                          at::Tensor w, at::Tensor mean,
-                                         at::Tensor rstd, int64_t n_stripes) {
+                                         at::Tensor rstd, int64_t n_stripes,
+                                         c10::optional<at::Tensor> dh) {
   CHECK_IN(dy); CHECK_IN(x); CHECK_IN(w);
   (void)n_stripes;  // stripe count is chosen device-side for full occupancy
   const int N = x.size(-1);
@@ -111,7 +126,13 @@ std::vector<at::Tensor> layernorm_bwd_dx(at::Tensor dy, at::Tensor x,
   auto pdb = at::empty({G, N}, f32);
   auto meanc = mean.contiguous();
   auto rstdc = rstd.contiguous();
-  check_hip(tdsa_ln_bwd_dx(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+  at::Tensor dhc;
+  void* dh_p = nullptr;
+  if (dh.has_value()) {
+    dhc = dh->contiguous();
+    dh_p = dhc.data_ptr();
+  }
+  check_hip(tdsa_ln_bwd_dx(dy.data_ptr(), dh_p, x.data_ptr(), w.data_ptr(),
                            meanc.data_ptr<float>(), rstdc.data_ptr<float>(),
                            dx.data_ptr(), pdw.data_ptr<float>(),
                            pdb.data_ptr<float>(), (int)M, N, dtype_flag(x),
@@ -415,8 +436,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("dbg_mfma", &dbg_mfma);
   mod.def("dbg_stage", &dbg_stage);
   mod.def("dbg_tr16", &dbg_tr16);
-  mod.def("layernorm_fwd", &layernorm_fwd);
-  mod.def("layernorm_bwd_dx", &layernorm_bwd_dx);
+  mod.def("layernorm_fwd", &layernorm_fwd, py::arg("x"), py::arg("w"),
+          py::arg("b"), py::arg("eps"), py::arg("res") = py::none());
+  mod.def("layernorm_bwd_dx", &layernorm_bwd_dx, py::arg("dy"), py::arg("x"),
+          py::arg("w"), py::arg("mean"), py::arg("rstd"),
+          py::arg("n_stripes"), py::arg("dh") = py::none());
   mod.def("layernorm_bwd_dwdb", &layernorm_bwd_dwdb);
   mod.def("gelu_fwd", &gelu_fwd);
   mod.def("gelu_bwd", &gelu_bwd);

@@ -30,8 +30,12 @@ template <> struct VecTraits<bf16> {
   static DEV_INLINE void store(bf16* p, V v) { store8(p, v); }
 };
 
-template <typename T>
-__global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+// HAS_RES: fused residual add — h = x + res is computed in-kernel, written
+// out (the residual stream) and normalized, saving the separate elementwise
+// add pass + launch per LayerNorm site (2 per transformer block).
+template <typename T, bool HAS_RES>
+__global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res,
+                              T* __restrict__ h, const T* __restrict__ w,
                               const T* __restrict__ b, T* __restrict__ y,
                               float* __restrict__ mean, float* __restrict__ rstd,
                               int M, int N, float eps) {
@@ -42,9 +46,20 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
   const int nth = blockDim.x;
   for (int row = blockIdx.x; row < M; row += gridDim.x) {
     const T* xr = x + (long long)row * N;
+    const T* rr = HAS_RES ? res + (long long)row * N : nullptr;
+    T* hr = HAS_RES ? h + (long long)row * N : nullptr;
     float s = 0.f, ss = 0.f;
     for (int i = tid * W; i < N; i += nth * W) {
       typename VT::V v = VT::load(xr + i);
+      if (HAS_RES) {
+        typename VT::V rv = VT::load(rr + i);
+        typename VT::V hv;
+#pragma unroll
+        for (int k = 0; k < W; ++k)
+          VT::set(hv, k, VT::get(v, k) + VT::get(rv, k));
+        VT::store(hr + i, hv);
+        v = hv;
+      }
 #pragma unroll
       for (int k = 0; k < W; ++k) {
         float f = VT::get(v, k);
@@ -60,9 +75,10 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
       mean[row] = mu;
       rstd[row] = rs;
     }
+    const T* inr = HAS_RES ? hr : xr;
     T* yr = y + (long long)row * N;
     for (int i = tid * W; i < N; i += nth * W) {
-      typename VT::V xv = VT::load(xr + i);
+      typename VT::V xv = VT::load(inr + i);
       typename VT::V wv = VT::load(w + i);
       typename VT::V bv = VT::load(b + i);
       typename VT::V ov;
@@ -79,8 +95,11 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
 // Backward dx + per-block dw/db stripe partials. Grid = G blocks; block g
 // handles rows g, g+G, ... and accumulates its dw/db into pdw[g*N..], so the
 // stripe write needs no atomics at all.
-template <typename T, int MAXITER>
-__global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__ x,
+// HAS_DH: dx += dh (gradient of the residual-stream output h), fusing the
+// backward-side elementwise add of the residual connection.
+template <typename T, int MAXITER, bool HAS_DH>
+__global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__ dh,
+                                 const T* __restrict__ x,
                                  const T* __restrict__ w, const float* __restrict__ mean,
                                  const float* __restrict__ rstd, T* __restrict__ dx,
                                  float* __restrict__ pdw, float* __restrict__ pdb,
@@ -125,17 +144,22 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__
     c1 /= N;
     c2 /= N;
     T* dxr = dx + (long long)row * N;
+    const T* dhr = HAS_DH ? dh + (long long)row * N : nullptr;
     for (int i = tid * W; i < N; i += nth * W) {
       typename VT::V dv = VT::load(dyr + i);
       typename VT::V xv = VT::load(xr + i);
       typename VT::V wv = VT::load(w + i);
+      typename VT::V hv;
+      if (HAS_DH) hv = VT::load(dhr + i);
       typename VT::V ov;
 #pragma unroll
       for (int k = 0; k < W; ++k) {
         float d = VT::get(dv, k);
         float xh = (VT::get(xv, k) - mu) * rs;
         float wdy = VT::get(wv, k) * d;
-        VT::set(ov, k, (wdy - (xh * c1 + c2)) * rs);
+        float g = (wdy - (xh * c1 + c2)) * rs;
+        if (HAS_DH) g += VT::get(hv, k);
+        VT::set(ov, k, g);
       }
       VT::store(dxr + i, ov);
     }
@@ -185,22 +209,25 @@ __global__ void ln_bwd_dwdb_kernel(const float* __restrict__ pdw,
 
 extern "C" {
 
-hipError_t tdsa_ln_fwd(const void* x, const void* w, const void* b, void* y,
-                       float* mean, float* rstd, int M, int N, float eps,
-                       int is_bf16, hipStream_t stream) {
+// res/h: optional fused residual (pass nullptr for the plain form).
+hipError_t tdsa_ln_fwd(const void* x, const void* res, void* h, const void* w,
+                       const void* b, void* y, float* mean, float* rstd,
+                       int M, int N, float eps, int is_bf16,
+                       hipStream_t stream) {
   const int block = 256;
   const int grid = (M < 4096) ? M : 4096;
+#define LN_FWD(T, HASR)                                                       \
+  hipLaunchKernelGGL((ln_fwd_kernel<T, HASR>), dim3(grid), dim3(block), 0,    \
+                     stream, (const T*)x, (const T*)res, (T*)h, (const T*)w,  \
+                     (const T*)b, (T*)y, mean, rstd, M, N, eps)
   if (is_bf16) {
     if (N % 8) return hipErrorInvalidValue;
-    hipLaunchKernelGGL(ln_fwd_kernel<bf16>, dim3(grid), dim3(block), 0, stream,
-                       (const bf16*)x, (const bf16*)w, (const bf16*)b, (bf16*)y,
-                       mean, rstd, M, N, eps);
+    if (res) LN_FWD(bf16, true); else LN_FWD(bf16, false);
   } else {
     if (N % 4) return hipErrorInvalidValue;
-    hipLaunchKernelGGL(ln_fwd_kernel<float>, dim3(grid), dim3(block), 0, stream,
-                       (const float*)x, (const float*)w, (const float*)b, (float*)y,
-                       mean, rstd, M, N, eps);
+    if (res) LN_FWD(float, true); else LN_FWD(float, false);
   }
+#undef LN_FWD
   return hipGetLastError();
 }
 
@@ -211,23 +238,25 @@ int tdsa_ln_bwd_dx_stripes(int M) {
   return g < 1 ? 1 : g;
 }
 
-hipError_t tdsa_ln_bwd_dx(const void* dy, const void* x, const void* w,
-                          const float* mean, const float* rstd, void* dx,
-                          float* pdw, float* pdb, int M, int N, int is_bf16,
-                          hipStream_t stream) {
+// dh: optional residual-stream gradient added into dx (nullptr for plain).
+hipError_t tdsa_ln_bwd_dx(const void* dy, const void* dh, const void* x,
+                          const void* w, const float* mean, const float* rstd,
+                          void* dx, float* pdw, float* pdb, int M, int N,
+                          int is_bf16, hipStream_t stream) {
   const int block = 256;
   const int grid = tdsa_ln_bwd_dx_stripes(M);
+#define LN_BWD(T, HASD)                                                       \
+  hipLaunchKernelGGL((ln_bwd_dx_kernel<T, 2, HASD>), dim3(grid), dim3(block), \
+                     0, stream, (const T*)dy, (const T*)dh, (const T*)x,      \
+                     (const T*)w, mean, rstd, (T*)dx, pdw, pdb, M, N)
   if (is_bf16) {
     if (N % 8 || N > 2 * block * 8) return hipErrorInvalidValue;
-    hipLaunchKernelGGL((ln_bwd_dx_kernel<bf16, 2>), dim3(grid), dim3(block), 0,
-                       stream, (const bf16*)dy, (const bf16*)x, (const bf16*)w,
-                       mean, rstd, (bf16*)dx, pdw, pdb, M, N);
+    if (dh) LN_BWD(bf16, true); else LN_BWD(bf16, false);
   } else {
     if (N % 4 || N > 2 * block * 4) return hipErrorInvalidValue;
-    hipLaunchKernelGGL((ln_bwd_dx_kernel<float, 2>), dim3(grid), dim3(block), 0,
-                       stream, (const float*)dy, (const float*)x, (const float*)w,
-                       mean, rstd, (float*)dx, pdw, pdb, M, N);
+    if (dh) LN_BWD(float, true); else LN_BWD(float, false);
   }
+#undef LN_BWD
   return hipGetLastError();
 }
 

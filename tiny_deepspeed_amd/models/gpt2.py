@@ -121,10 +121,26 @@ class Block(nn.Module):
         self.ln_2 = nn.LayerNorm(config.n_embd)
         self.mlp = MLP(config)
 
-    def forward(self, x):
-        x = x + self.attn(self.ln_1(x))
-        x = x + self.mlp(self.ln_2(x))
-        return x
+    def forward(self, x, delta=None):
+        """Residual stream threading: takes (x, pending delta) and returns
+        (h, pending delta) so each residual add fuses into the next
+        LayerNorm kernel (wrapped modules expose forward_fused; the raw
+        nn.LayerNorm path does the adds explicitly — same math)."""
+        if delta is None:
+            y1 = self.ln_1(x)
+            h = x
+        elif hasattr(self.ln_1, "forward_fused"):
+            h, y1 = self.ln_1.forward_fused(x, delta)
+        else:
+            h = x + delta
+            y1 = self.ln_1(h)
+        a = self.attn(y1)
+        if hasattr(self.ln_2, "forward_fused"):
+            h2, y2 = self.ln_2.forward_fused(h, a)
+        else:
+            h2 = h + a
+            y2 = self.ln_2(h2)
+        return h2, self.mlp(y2)
 
 
 class GPT2Model(nn.Module):
@@ -159,10 +175,17 @@ class GPT2Model(nn.Module):
         x = tok + posemb
         if self.config.dropout > 0:
             x = nn.functional.dropout(x, self.config.dropout, self.training)
+        delta = None
         for block in self.transformer.h:
-            x = block(x)
-        x = self.transformer.ln_f(x)
-        logits = self.lm_head(x)
+            x, delta = block(x, delta)
+        ln_f = self.transformer.ln_f
+        if delta is None:
+            y = ln_f(x)
+        elif hasattr(ln_f, "forward_fused"):
+            _, y = ln_f.forward_fused(x, delta)
+        else:
+            y = ln_f(x + delta)
+        logits = self.lm_head(y)
         loss = None
         if targets is not None:
             loss = ops.cross_entropy(logits, targets)

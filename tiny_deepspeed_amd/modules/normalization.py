@@ -12,6 +12,24 @@ import torch.nn as nn
 from .. import ops
 
 
+class _LayerNormResFn(torch.autograd.Function):
+    """Fused residual-add + LayerNorm: returns (h, y) with h = x + res.
+    Backward folds the residual-stream gradient dh into dx (one kernel)."""
+
+    @staticmethod
+    def forward(ctx, x, res, weight, bias, module):
+        ctx.module = module
+        h, y, mean, rstd = module.forward_res_callback(x, res, weight, bias)
+        ctx.save_for_backward(h, mean, rstd)
+        return h, y
+
+    @staticmethod
+    def backward(ctx, dh, dy):
+        h, mean, rstd = ctx.saved_tensors
+        dx, dw, db = ctx.module.backward_callback(dy, h, mean, rstd, dh=dh)
+        return dx, dx, dw, db, None
+
+
 class _LayerNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, module):
@@ -49,8 +67,13 @@ class LayerNorm(nn.LayerNorm):
     def forward_callback(self, x, weight, bias):
         return ops.layernorm_fwd(x, weight, bias, eps=self.eps, tuner=self.tuner)
 
-    def backward_callback(self, dy, x, mean, rstd):
-        dx, ws = ops.layernorm_dx(dy, x, self.weight, mean, rstd, tuner=self.tuner)
+    def forward_res_callback(self, x, res, weight, bias):
+        return ops.layernorm_fwd_res(x, res, weight, bias, eps=self.eps,
+                                     tuner=self.tuner)
+
+    def backward_callback(self, dy, x, mean, rstd, dh=None):
+        dx, ws = ops.layernorm_dx(dy, x, self.weight, mean, rstd, dh=dh,
+                                  tuner=self.tuner)
         if self.weight.requires_grad:
             dw, db = ops.layernorm_dwdb(ws, dtype=self.weight.dtype, tuner=self.tuner)
             self._assert_grad_shapes(dw, db)
@@ -66,3 +89,7 @@ class LayerNorm(nn.LayerNorm):
 
     def forward(self, x):
         return _LayerNormFn.apply(x, self.weight, self.bias, self)
+
+    def forward_fused(self, x, res):
+        """(h, y) = (x + res, LN(x + res)) with the add fused in-kernel."""
+        return _LayerNormResFn.apply(x, res, self.weight, self.bias, self)

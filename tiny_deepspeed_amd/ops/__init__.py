@@ -12,7 +12,8 @@ from .linear import (
     linear_weight_grad,
     linear_bias_grad,
 )
-from .layernorm import layernorm_fwd, layernorm_dx, layernorm_dwdb
+from .layernorm import (layernorm_fwd, layernorm_fwd_res, layernorm_dx,
+                        layernorm_dwdb)
 from .embedding import embedding_forward, embedding_weight_grad
 from .gelu import gelu, gelu_fwd, gelu_bwd
 from .attention import causal_attention, fused_causal_attention
@@ -24,7 +25,7 @@ from .utils import acc_dtype
 
 __all__ = [
     "linear_forward", "linear_input_grad", "linear_weight_grad", "linear_bias_grad",
-    "layernorm_fwd", "layernorm_dx", "layernorm_dwdb",
+    "layernorm_fwd", "layernorm_fwd_res", "layernorm_dx", "layernorm_dwdb",
     "embedding_forward", "embedding_weight_grad",
     "gelu", "gelu_fwd", "gelu_bwd",
     "causal_attention", "fused_causal_attention",

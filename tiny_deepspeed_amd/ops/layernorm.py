@@ -38,13 +38,29 @@ def layernorm_fwd(x, weight, bias, eps=1e-5, tuner=None):
     return y.to(x.dtype), mean, rstd
 
 
-def layernorm_dx(dy, x, weight, mean, rstd, tuner=None):
-    """Input gradient. On GPU this also produces stripe partials for dw/db
-    (returned as an opaque workspace consumed by layernorm_dwdb)."""
+def layernorm_fwd_res(x, res, weight, bias, eps=1e-5, tuner=None):
+    """Fused residual + LayerNorm: h = x + res; y = LN(h).
+    Returns (h, y, mean, rstd) — h is the residual stream, written once by
+    the kernel instead of a separate elementwise add pass."""
+    if _ext.use_native(x):
+        y, mean, rstd, h = _ext.get_ext().layernorm_fwd(
+            x.contiguous(), weight, bias, eps, res.contiguous()
+        )
+        return h, y, mean, rstd
+    h = x + res
+    y, mean, rstd = layernorm_fwd(h, weight, bias, eps)
+    return h, y, mean, rstd
+
+
+def layernorm_dx(dy, x, weight, mean, rstd, dh=None, tuner=None):
+    """Input gradient (+ optional fused add of the residual-stream grad dh).
+    On GPU this also produces stripe partials for dw/db (returned as an
+    opaque workspace consumed by layernorm_dwdb)."""
     if _ext.use_native(dy):
         ext = _ext.get_ext()
         dx, pdw, pdb = ext.layernorm_bwd_dx(
-            dy.contiguous(), x.contiguous(), weight, mean, rstd, N_STRIPES
+            dy.contiguous(), x.contiguous(), weight, mean, rstd, N_STRIPES,
+            dh,
         )
         return dx, (pdw, pdb)
     xf = x.float()
@@ -55,6 +71,8 @@ def layernorm_dx(dy, x, weight, mean, rstd, tuner=None):
     c1 = (xhat * wdy).mean(dim=-1, keepdim=True)
     c2 = wdy.mean(dim=-1, keepdim=True)
     dx = (wdy - (xhat * c1 + c2)) * rstd.unsqueeze(-1)
+    if dh is not None:
+        dx = dx + dh.float()
     return dx.to(x.dtype), (dy, x, mean, rstd)
 
 

@@ -36,8 +36,9 @@ class Linear(base.Linear):
 class LayerNorm(base.LayerNorm):
     _mode = ALLREDUCE
 
-    def backward_callback(self, dy, x, mean, rstd):
-        dx, ws = ops.layernorm_dx(dy, x, self.weight, mean, rstd, tuner=self.tuner)
+    def backward_callback(self, dy, x, mean, rstd, dh=None):
+        dx, ws = ops.layernorm_dx(dy, x, self.weight, mean, rstd, dh=dh,
+                                  tuner=self.tuner)
         if self.weight.requires_grad:
             dw, db = ops.layernorm_dwdb(ws, dtype=self.weight.dtype, tuner=self.tuner)
             publish_grad(self._comm, self.weight, dw, self._mode)

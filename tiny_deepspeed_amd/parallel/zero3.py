@@ -111,9 +111,15 @@ class LayerNorm(_GatherMixin, base.LayerNorm):
         return ops.layernorm_fwd(x, bufs["weight"], bufs["bias"],
                                  eps=self.eps, tuner=self.tuner)
 
-    def backward_callback(self, dy, x, mean, rstd):
+    def forward_res_callback(self, x, res, weight, bias):
+        bufs = self._take_bufs("_tdsa_next")
+        return ops.layernorm_fwd_res(x, res, bufs["weight"], bufs["bias"],
+                                     eps=self.eps, tuner=self.tuner)
+
+    def backward_callback(self, dy, x, mean, rstd, dh=None):
         w = self._take_bufs("_tdsa_prev")["weight"]
-        dx, ws = ops.layernorm_dx(dy, x, w, mean, rstd, tuner=self.tuner)
+        dx, ws = ops.layernorm_dx(dy, x, w, mean, rstd, dh=dh,
+                                  tuner=self.tuner)
         if self.weight.requires_grad:
             dw, db = ops.layernorm_dwdb(ws, dtype=self.weight.dtype,
                                         tuner=self.tuner)
