@@ -11,6 +11,8 @@ from collections import OrderedDict
 
 import torch
 
+from ..utils.profiling import trace_range
+
 
 class Optimizer:
     def __init__(self, parameters, lr):
@@ -39,6 +41,11 @@ class Optimizer:
     @torch.no_grad()
     def step(self):
         self.t += 1
+        with trace_range("tdsa:optimizer.step"):
+            self._run_step()
+
+    @torch.no_grad()
+    def _run_step(self):
         self.pre_step()
         self._apply_updates(
             [(n, p) for n, p in self.params.items()

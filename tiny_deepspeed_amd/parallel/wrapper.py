@@ -21,6 +21,7 @@ import torch
 import torch.nn as nn
 
 from .. import modules as base_modules
+from ..utils.profiling import trace_range
 from .comm import CommContext, default_comm
 
 
@@ -133,7 +134,8 @@ class ModelWrapper(nn.Module):
         if self.require_backward_grad_sync:
             for p in self.module.parameters():
                 arm(p)
-        return self.module(*args, **kwargs)
+        with trace_range("tdsa:forward"):
+            return self.module(*args, **kwargs)
 
     def named_parameters(self, *a, **k):
         return self.module.named_parameters(*a, **k)
