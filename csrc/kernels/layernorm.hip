@@ -40,6 +40,7 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res
                               int M, int N, float eps) {
   using VT = VecTraits<T>;
   constexpr int W = VT::W;
+  constexpr int MAXITER = 2;  // rows cached in registers between passes
   __shared__ float scratch[2 * 1024 / WAVE];
   const int tid = threadIdx.x;
   const int nth = blockDim.x;
@@ -47,23 +48,28 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res
     const T* xr = x + (long long)row * N;
     const T* rr = HAS_RES ? res + (long long)row * N : nullptr;
     T* hr = HAS_RES ? h + (long long)row * N : nullptr;
+    typename VT::V xc[MAXITER];
     float s = 0.f, ss = 0.f;
-    for (int i = tid * W; i < N; i += nth * W) {
-      typename VT::V v = VT::load(xr + i);
-      if (HAS_RES) {
-        typename VT::V rv = VT::load(rr + i);
-        typename VT::V hv;
+    {
+      int it = 0;
+      for (int i = tid * W; i < N; i += nth * W, ++it) {
+        typename VT::V v = VT::load(xr + i);
+        if (HAS_RES) {
+          typename VT::V rv = VT::load(rr + i);
+          typename VT::V hv;
 #pragma unroll
-        for (int k = 0; k < W; ++k)
-          VT::set(hv, k, VT::get(v, k) + VT::get(rv, k));
-        VT::store(hr + i, hv);
-        v = hv;
-      }
+          for (int k = 0; k < W; ++k)
+            VT::set(hv, k, VT::get(v, k) + VT::get(rv, k));
+          VT::store(hr + i, hv);
+          v = hv;
+        }
+        xc[it] = v;
 #pragma unroll
-      for (int k = 0; k < W; ++k) {
-        float f = VT::get(v, k);
-        s += f;
-        ss += f * f;
+        for (int k = 0; k < W; ++k) {
+          float f = VT::get(v, k);
+          s += f;
+          ss += f * f;
+        }
       }
     }
     block_sum2(s, ss, scratch);
@@ -74,19 +80,21 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res
       mean[row] = mu;
       rstd[row] = rs;
     }
-    const T* inr = HAS_RES ? hr : xr;
     T* yr = y + (long long)row * N;
-    for (int i = tid * W; i < N; i += nth * W) {
-      typename VT::V xv = VT::load(inr + i);
-      typename VT::V wv = VT::load(w + i);
-      typename VT::V bv = VT::load(b + i);
-      typename VT::V ov;
+    {
+      int it = 0;
+      for (int i = tid * W; i < N; i += nth * W, ++it) {
+        typename VT::V xv = xc[it];
+        typename VT::V wv = VT::load(w + i);
+        typename VT::V bv = VT::load(b + i);
+        typename VT::V ov;
 #pragma unroll
-      for (int k = 0; k < W; ++k) {
-        float xh = (VT::get(xv, k) - mu) * rs;
-        VT::set(ov, k, xh * VT::get(wv, k) + VT::get(bv, k));
+        for (int k = 0; k < W; ++k) {
+          float xh = (VT::get(xv, k) - mu) * rs;
+          VT::set(ov, k, xh * VT::get(wv, k) + VT::get(bv, k));
+        }
+        VT::store(yr + i, ov);
       }
-      VT::store(yr + i, ov);
     }
   }
 }
@@ -120,6 +128,7 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__
     const T* xr = x + (long long)row * N;
     const float mu = mean[row];
     const float rs = rstd[row];
+    typename VT::V dyc[MAXITER], xc[MAXITER];
     float c1 = 0.f, c2 = 0.f;
     {
       int it = 0;
@@ -127,6 +136,8 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__
         typename VT::V dv = VT::load(dyr + i);
         typename VT::V xv = VT::load(xr + i);
         typename VT::V wv = VT::load(w + i);
+        dyc[it] = dv;
+        xc[it] = xv;
 #pragma unroll
         for (int k = 0; k < W; ++k) {
           float d = VT::get(dv, k);
@@ -144,23 +155,26 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__
     c2 /= N;
     T* dxr = dx + (long long)row * N;
     const T* dhr = HAS_DH ? dh + (long long)row * N : nullptr;
-    for (int i = tid * W; i < N; i += nth * W) {
-      typename VT::V dv = VT::load(dyr + i);
-      typename VT::V xv = VT::load(xr + i);
-      typename VT::V wv = VT::load(w + i);
-      typename VT::V hv;
-      if (HAS_DH) hv = VT::load(dhr + i);
-      typename VT::V ov;
+    {
+      int it = 0;
+      for (int i = tid * W; i < N; i += nth * W, ++it) {
+        typename VT::V dv = dyc[it];
+        typename VT::V xv = xc[it];
+        typename VT::V wv = VT::load(w + i);
+        typename VT::V hv;
+        if (HAS_DH) hv = VT::load(dhr + i);
+        typename VT::V ov;
 #pragma unroll
-      for (int k = 0; k < W; ++k) {
-        float d = VT::get(dv, k);
-        float xh = (VT::get(xv, k) - mu) * rs;
-        float wdy = VT::get(wv, k) * d;
-        float g = (wdy - (xh * c1 + c2)) * rs;
-        if (HAS_DH) g += VT::get(hv, k);
-        VT::set(ov, k, g);
+        for (int k = 0; k < W; ++k) {
+          float d = VT::get(dv, k);
+          float xh = (VT::get(xv, k) - mu) * rs;
+          float wdy = VT::get(wv, k) * d;
+          float g = (wdy - (xh * c1 + c2)) * rs;
+          if (HAS_DH) g += VT::get(hv, k);
+          VT::set(ov, k, g);
+        }
+        VT::store(dxr + i, ov);
       }
-      VT::store(dxr + i, ov);
     }
   }
   // stripe write: block-owned rows of pdw/pdb
@@ -220,10 +234,10 @@ hipError_t tdsa_ln_fwd(const void* x, const void* res, void* h, const void* w,
                      stream, (const T*)x, (const T*)res, (T*)h, (const T*)w,  \
                      (const T*)b, (T*)y, mean, rstd, M, N, eps)
   if (is_bf16) {
-    if (N % 8) return hipErrorInvalidValue;
+    if (N % 8 || N > 2 * block * 8) return hipErrorInvalidValue;
     if (res) LN_FWD(bf16, true); else LN_FWD(bf16, false);
   } else {
-    if (N % 4) return hipErrorInvalidValue;
+    if (N % 4 || N > 2 * block * 4) return hipErrorInvalidValue;
     if (res) LN_FWD(float, true); else LN_FWD(float, false);
   }
 #undef LN_FWD
