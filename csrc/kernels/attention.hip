@@ -355,21 +355,31 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
 // ---------------------------------------------------------------------------
 // Backward: delta = rowsum(dO * O)
 // ---------------------------------------------------------------------------
-// grid: (ceil(T/waves_per_block), B*H) — no per-thread 64-bit div/mod (a
-// 1-D flat-index version spent ~234 VALU/row on the divisions).
+// grid: (ceil(T/rows_per_block), B*H); 8 rows per wave, 8 bf16 per lane
+// (a one-element-per-lane version loaded 2 B per lane and ran at 1.5 TB/s).
 __global__ void attn_delta_kernel(const bf16* __restrict__ dout,
                                   const bf16* __restrict__ o,
                                   float* __restrict__ delta, int T, int H,
                                   GStride so) {
-  const int t = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+  constexpr int ROWS_PER_WAVE = 8;  // D=64 = 8 lanes x 8 elements
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int t = (blockIdx.x * (blockDim.x / WAVE) + wid) * ROWS_PER_WAVE
+                + (lane >> 3);
   if (t >= T) return;
   const long long bh = blockIdx.y;
-  const int lane = threadIdx.x & (WAVE - 1);
   const long long off = (bh / H) * so.b + (bh % H) * so.h
-                        + (long long)t * so.t + lane;
-  float acc = bf2f(dout[off]) * bf2f(o[off]);
-  acc = wave_sum(acc);
-  if (lane == 0) delta[bh * T + t] = acc;
+                        + (long long)t * so.t + (lane & 7) * 8;
+  short8v dv = load8(dout + off);
+  short8v ov = load8(o + off);
+  float acc = 0.f;
+#pragma unroll
+  for (int k = 0; k < 8; ++k) acc += bf_elem(dv, k) * bf_elem(ov, k);
+  // reduce across the 8 lanes sharing a row
+  acc += __shfl_xor(acc, 1, WAVE);
+  acc += __shfl_xor(acc, 2, WAVE);
+  acc += __shfl_xor(acc, 4, WAVE);
+  if ((lane & 7) == 0) delta[bh * T + t] = acc;
 }
 
 // ---------------------------------------------------------------------------
@@ -681,7 +691,7 @@ hipError_t tdsa_attn_bwd(const void* q, const void* k, const void* v,
   GStride sd{sd_in[0], sd_in[1], (int)sd_in[2]};
   const long long BH = B * H;
   {
-    const int rows_per_block = 256 / WAVE;
+    const int rows_per_block = (256 / WAVE) * 8;
     hipLaunchKernelGGL(attn_delta_kernel,
                        dim3((T + rows_per_block - 1) / rows_per_block, BH),
                        dim3(256), 0, stream, (const bf16*)dout, (const bf16*)o,

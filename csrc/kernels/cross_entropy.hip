@@ -9,12 +9,14 @@
 
 namespace {
 
-// merge two (m, s) logsumexp states
+// merge two (m, s) logsumexp states (raw v_exp: exp(-inf) == 0, and the
+// branchless form keeps both scales well-defined for m == m2 == -inf)
 DEV_INLINE void lse_merge(float& m, float& s, float m2, float s2) {
   float mn = fmaxf(m, m2);
-  // exp(-inf - -inf) guard: if both -inf keep s at 0
-  float a = (m == -INFINITY) ? 0.f : __expf(m - mn);
-  float b = (m2 == -INFINITY) ? 0.f : __expf(m2 - mn);
+  float a = (m == -INFINITY) ? 0.f
+            : __builtin_amdgcn_exp2f(1.4426950408889634f * (m - mn));
+  float b = (m2 == -INFINITY) ? 0.f
+            : __builtin_amdgcn_exp2f(1.4426950408889634f * (m2 - mn));
   s = s * a + s2 * b;
   m = mn;
 }
@@ -54,18 +56,20 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
   for (long long row = blockIdx.x; row < R; row += gridDim.x) {
     const T* lr = logits + row * V;
     float m = -INFINITY, s = 0.f;
+    // per-CHUNK online merge: a per-element data-dependent branch diverges
+    // and serializes; instead reduce the chunk max, sum exps against it,
+    // and merge (m, s) once per chunk.
     for (int i = tid * W; i + W <= V; i += blockDim.x * W) {
       typename RV::V v = RV::load(lr + i);
+      float cm = RV::get(v, 0);
 #pragma unroll
-      for (int k = 0; k < W; ++k) {
-        float f = RV::get(v, k);
-        if (f > m) {
-          s = s * __expf(m - f) + 1.0f;
-          m = f;
-        } else {
-          s += __expf(f - m);
-        }
-      }
+      for (int k = 1; k < W; ++k) cm = fmaxf(cm, RV::get(v, k));
+      float cs = 0.f;
+#pragma unroll
+      for (int k = 0; k < W; ++k)
+        cs += __builtin_amdgcn_exp2f(1.4426950408889634f
+                                     * (RV::get(v, k) - cm));
+      lse_merge(m, s, cm, cs);
     }
     const int rem0 = (V / W) * W;
     for (int j = rem0 + tid; j < V; j += blockDim.x) {
