@@ -83,32 +83,38 @@ struct GStride {
   int t;
 };
 
-// Per-lane staging addresses for one [64][64] tile: 512 chunks of 8 bf16.
-struct TileAddr {
-  int stage_src[4];
-  int stage_dst[4];
-};
-
+// Per-lane staging state for [64][64] tiles: absolute per-thread source
+// pointers (bumped by a constant per tile — recomputing the 64-bit address
+// per chunk per tile was ~70 VALU/tile in the disassembly) + swizzled LDS
+// byte offsets. 512 chunks of 8 bf16 over NT threads.
 template <int NT>
-DEV_INLINE TileAddr tile_addr(int tid, int t_stride) {
-  TileAddr a;
+struct Stage {
+  static constexpr int REPS = 512 / NT;
+  const bf16* src[REPS];
+  int dst[REPS];
+  int step;  // elements to advance per tile
+
+  DEV_INLINE Stage(const bf16* g, int tid, int t_stride) {
+    step = KVB * t_stride;
 #pragma unroll
-  for (int rep = 0; rep < 512 / NT * 8 / 8; ++rep) {
-    int chunk = tid + rep * NT;
-    a.stage_src[rep] = (chunk >> 3) * t_stride + (chunk & 7) * 8;
-    a.stage_dst[rep] = swz(chunk >> 3, (chunk & 7) * 16);
+    for (int rep = 0; rep < REPS; ++rep) {
+      int chunk = tid + rep * NT;
+      src[rep] = g + (chunk >> 3) * t_stride + (chunk & 7) * 8;
+      dst[rep] = swz(chunk >> 3, (chunk & 7) * 16);
+    }
   }
-  return a;
-}
 
-template <int NT>
-DEV_INLINE void stage_rowmajor(const bf16* __restrict__ g, char* lds,
-                               const TileAddr& a) {
+  DEV_INLINE void run(char* lds) {
 #pragma unroll
-  for (int rep = 0; rep < 512 / NT; ++rep)
-    *reinterpret_cast<short8v*>(lds + a.stage_dst[rep]) =
-        load8(g + a.stage_src[rep]);
-}
+    for (int rep = 0; rep < REPS; ++rep)
+      *reinterpret_cast<short8v*>(lds + dst[rep]) = load8(src[rep]);
+  }
+
+  DEV_INLINE void advance() {
+#pragma unroll
+    for (int rep = 0; rep < REPS; ++rep) src[rep] += step;
+  }
+};
 
 // C-layout -> A-fragment repack. Values live per lane as 16 C registers per
 // 32-wide tile (packed to bf16 word pairs wA[r1]=(r0=0,1), wB[r1]=(r0=2,3)).
@@ -220,7 +226,8 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
   const bf16* kp = k + boff;
   const bf16* vp = v + boff;
 
-  const TileAddr ta = tile_addr<NT>(tid, sq.t);
+  Stage<NT> stage_k(kp, tid, sq.t);
+  Stage<NT> stage_v(vp, tid, sq.t);
   int kf_off[2][4];  // [32-tile][k-slice]
 #pragma unroll
   for (int t2 = 0; t2 < 2; ++t2)
@@ -245,8 +252,10 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
   const int n_kv = (qb + 1) * BM / KVB;
   for (int j = 0; j < n_kv; ++j) {
     __syncthreads();
-    stage_rowmajor<NT>(kp + (long long)(j * KVB) * sq.t, lds_k, ta);
-    stage_rowmajor<NT>(vp + (long long)(j * KVB) * sq.t, lds_v, ta);
+    stage_k.run(lds_k);
+    stage_v.run(lds_v);
+    stage_k.advance();
+    stage_v.advance();
     __syncthreads();
 
     const int key0 = j * KVB;
@@ -419,8 +428,8 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
   const bf16* vp = v + boff;
   const bf16* dop = dout + ooff;
 
-  const TileAddr ta = tile_addr<NT>(tid, sq.t);
-  const TileAddr tao = tile_addr<NT>(tid, so.t);
+  Stage<NT> stage_q(qp + (long long)(jb * BK / KVB * KVB) * sq.t, tid, sq.t);
+  Stage<NT> stage_do(dop + (long long)(jb * BK / KVB * KVB) * so.t, tid, so.t);
   int af_off[2][4];
 #pragma unroll
   for (int t2 = 0; t2 < 2; ++t2)
@@ -445,8 +454,10 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
 
   for (int i = jb * BK / KVB; i < T / KVB; ++i) {
     __syncthreads();
-    stage_rowmajor<NT>(qp + (long long)(i * KVB) * sq.t, lds_q, ta);
-    stage_rowmajor<NT>(dop + (long long)(i * KVB) * so.t, lds_do, tao);
+    stage_q.run(lds_q);
+    stage_do.run(lds_do);
+    stage_q.advance();
+    stage_do.advance();
     if (tid < KVB) {
       lds_lse[tid] = lse[bh * T + i * KVB + tid] * LOG2E;
       lds_dlt[tid] = delta[bh * T + i * KVB + tid];
@@ -562,7 +573,8 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   const bf16* vp = v + boff;
   const bf16* dop = dout + ooff + (long long)(qb * BM) * so.t;
 
-  const TileAddr ta = tile_addr<NT>(tid, sq.t);
+  Stage<NT> stage_k(kp, tid, sq.t);
+  Stage<NT> stage_v(vp, tid, sq.t);
   int f_off[2][4];
 #pragma unroll
   for (int t2 = 0; t2 < 2; ++t2)
@@ -589,8 +601,10 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   const int n_kv = (qb + 1) * BM / KVB;
   for (int j = 0; j < n_kv; ++j) {
     __syncthreads();
-    stage_rowmajor<NT>(kp + (long long)(j * KVB) * sq.t, lds_k, ta);
-    stage_rowmajor<NT>(vp + (long long)(j * KVB) * sq.t, lds_v, ta);
+    stage_k.run(lds_k);
+    stage_v.run(lds_v);
+    stage_k.advance();
+    stage_v.advance();
     __syncthreads();
 
     const int key0 = j * KVB;
