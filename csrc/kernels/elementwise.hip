@@ -10,12 +10,22 @@ namespace {
 
 constexpr float GC0 = 0.7978845608028654f;  // sqrt(2/pi)
 constexpr float GC1 = 0.044715f;
+constexpr float TWO_LOG2E = 2.8853900817779268f;  // 2*log2(e)
+
+// tanh via raw v_exp/v_rcp: tanh(t) = 1 - 2/(exp(2t)+1). Saturates cleanly
+// (exp2(+inf)=inf -> 1, exp2(-inf)=0 -> -1); ~2-3 ulp, inside bf16/fp32
+// tolerance for GELU. libm tanhf is a long polynomial/branch chain and made
+// the elementwise kernels VALU-bound.
+DEV_INLINE float fast_tanh(float t) {
+  float e = __builtin_amdgcn_exp2f(TWO_LOG2E * t);
+  return 1.0f - 2.0f * __builtin_amdgcn_rcpf(e + 1.0f);
+}
 
 DEV_INLINE float gelu_f(float x) {
-  return 0.5f * x * (1.0f + tanhf(GC0 * (x + GC1 * x * x * x)));
+  return 0.5f * x * (1.0f + fast_tanh(GC0 * (x + GC1 * x * x * x)));
 }
 DEV_INLINE float gelu_df(float x) {
-  float t = tanhf(GC0 * (x + GC1 * x * x * x));
+  float t = fast_tanh(GC0 * (x + GC1 * x * x * x));
   float dt = (1.0f - t * t) * GC0 * (1.0f + 3.0f * GC1 * x * x);
   return 0.5f * (1.0f + t) + 0.5f * x * dt;
 }
