@@ -77,6 +77,19 @@ DEV_INLINE bfrag lds_read16(const char* lds, int byte_off) {
 // Row index of C register r for this half-wave (the 32x32 C layout).
 DEV_INLINE int crow(int r, int h32) { return (r & 3) + 8 * (r >> 2) + 4 * h32; }
 
+// XCD-aware block remap (T1): the dispatcher places linear block b on XCD
+// b % 8, and the x-dimension varies fastest, so the gridDim.x blocks that
+// share one (batch, head) — and its K/V tiles — land on different XCDs'
+// L2s. Remap so they share an XCD: within each window of 8*gridX linear
+// ids, ids congruent mod 8 form one bh. Bijective when gridY % 8 == 0.
+DEV_INLINE void xcd_remap(int gx, long long gy, int& tile, long long& bh) {
+  if (gy % 8 == 0) {
+    const long long id = bh * gx + tile;   // linear dispatch id
+    tile = (int)((id % (8 * gx)) / 8);
+    bh = (id / (8 * gx)) * 8 + (id % 8);
+  }
+}
+
 // Per-tensor global strides in ELEMENTS (last dim contiguous).
 struct GStride {
   long long b;
@@ -214,8 +227,9 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
   char* lds_k = smem;                 // [64][64] keys row-major
   char* lds_v = smem + KVB * D * 2;   // [64][64] values row-major
 
-  const int qb = blockIdx.x;
-  const long long bh = blockIdx.y;
+  int qb = blockIdx.x;
+  long long bh = blockIdx.y;
+  xcd_remap(gridDim.x, gridDim.y, qb, bh);
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int w = tid / WAVE;
@@ -414,8 +428,9 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
   float* lds_lse = reinterpret_cast<float*>(smem + 2 * KVB * D * 2);
   float* lds_dlt = lds_lse + KVB;
 
-  const int jb = blockIdx.x;
-  const long long bh = blockIdx.y;
+  int jb = blockIdx.x;
+  long long bh = blockIdx.y;
+  xcd_remap(gridDim.x, gridDim.y, jb, bh);
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int w = tid / WAVE;
@@ -559,8 +574,9 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   char* lds_k = smem;                     // K row-major
   char* lds_v = smem + KVB * D * 2;       // V row-major
 
-  const int qb = blockIdx.x;
-  const long long bh = blockIdx.y;
+  int qb = blockIdx.x;
+  long long bh = blockIdx.y;
+  xcd_remap(gridDim.x, gridDim.y, qb, bh);
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int w = tid / WAVE;
