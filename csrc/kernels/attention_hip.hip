@@ -128,6 +128,19 @@ struct Stage {
 #pragma unroll
     for (int rep = 0; rep < REPS; ++rep) src[rep] += step;
   }
+
+  // T14 split: issue the global loads early (they fly under the previous
+  // tile's compute), write to LDS later.
+  DEV_INLINE void fetch(short8v (&buf)[REPS]) {
+#pragma unroll
+    for (int rep = 0; rep < REPS; ++rep) buf[rep] = load8(src[rep]);
+  }
+
+  DEV_INLINE void put(char* lds, const short8v (&buf)[REPS]) {
+#pragma unroll
+    for (int rep = 0; rep < REPS; ++rep)
+      *reinterpret_cast<short8v*>(lds + dst[rep]) = buf[rep];
+  }
 };
 
 // C-layout -> A-fragment repack. Values live per lane as 16 C registers per
@@ -616,12 +629,19 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   f32x16 dq_acc[2] = {};
 
   const int n_kv = (qb + 1) * BM / KVB;
+  short8v pend_k[Stage<NT>::REPS], pend_v[Stage<NT>::REPS];
+  stage_k.fetch(pend_k);
+  stage_v.fetch(pend_v);
   for (int j = 0; j < n_kv; ++j) {
     __syncthreads();
-    stage_k.run(lds_k);
-    stage_v.run(lds_v);
-    stage_k.advance();
-    stage_v.advance();
+    stage_k.put(lds_k, pend_k);
+    stage_v.put(lds_v, pend_v);
+    if (j + 1 < n_kv) {
+      stage_k.advance();
+      stage_v.advance();
+      stage_k.fetch(pend_k);   // next tile flies under this tile's compute
+      stage_v.fetch(pend_v);
+    }
     __syncthreads();
 
     const int key0 = j * KVB;
