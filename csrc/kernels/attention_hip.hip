@@ -459,12 +459,12 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
 
   Stage<NT> stage_q(qp + (long long)(jb * BK / KVB * KVB) * sq.t, tid, sq.t);
   Stage<NT> stage_do(dop + (long long)(jb * BK / KVB * KVB) * so.t, tid, so.t);
-  int af_off[2][4];
-#pragma unroll
-  for (int t2 = 0; t2 < 2; ++t2)
-#pragma unroll
-    for (int s = 0; s < 4; ++s)
-      af_off[t2][s] = swz(t2 * 32 + k32, (s * 16 + 8 * h32) * 2);
+  // fragment offsets on the fly: af(t2, s) = rowbase[t2] + (s-term ^ xk);
+  // the s/h term and the swizzle XOR live in the same byte bits 4-6, so XOR
+  // composes (keeps 8 VGPRs free for the staging split below)
+  const int af_row[2] = {k32 * 128, (32 + k32) * 128};
+  const int af_xk = (k32 & 7) << 4;
+#define AF_OFF(t2, s) (af_row[t2] + (((s) * 32 + 16 * h32) ^ af_xk))
   int trb[2][2];
   tr_bases(lane, trb);
   const float kscale = scale * LOG2E;
@@ -481,12 +481,20 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
   f32x16 dk_acc[2] = {};
   f32x16 dv_acc[2] = {};
 
-  for (int i = jb * BK / KVB; i < T / KVB; ++i) {
+  const int i0 = jb * BK / KVB;
+  short8v pend_q[Stage<NT>::REPS], pend_do[Stage<NT>::REPS];
+  stage_q.fetch(pend_q);
+  stage_do.fetch(pend_do);
+  for (int i = i0; i < T / KVB; ++i) {
     __syncthreads();
-    stage_q.run(lds_q);
-    stage_do.run(lds_do);
-    stage_q.advance();
-    stage_do.advance();
+    stage_q.put(lds_q, pend_q);
+    stage_do.put(lds_do, pend_do);
+    if (i + 1 < T / KVB) {
+      stage_q.advance();
+      stage_do.advance();
+      stage_q.fetch(pend_q);   // next tile flies under this tile's compute
+      stage_do.fetch(pend_do);
+    }
     if (tid < KVB) {
       lds_lse[tid] = lse[bh * T + i * KVB + tid] * LOG2E;
       lds_dlt[tid] = delta[bh * T + i * KVB + tid];
@@ -506,8 +514,8 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int s = 0; s < 4; ++s) {
-        s_acc = MFMA32(lds_read16(lds_q, af_off[t2][s]), k_frag[s], s_acc);
-        dp_acc = MFMA32(lds_read16(lds_do, af_off[t2][s]), v_frag[s], dp_acc);
+        s_acc = MFMA32(lds_read16(lds_q, AF_OFF(t2, s)), k_frag[s], s_acc);
+        dp_acc = MFMA32(lds_read16(lds_do, AF_OFF(t2, s)), v_frag[s], dp_acc);
       }
       __builtin_amdgcn_s_setprio(0);
 #pragma unroll
