@@ -255,12 +255,10 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
 
   Stage<NT> stage_k(kp, tid, sq.t);
   Stage<NT> stage_v(vp, tid, sq.t);
-  int kf_off[2][4];  // [32-tile][k-slice]
-#pragma unroll
-  for (int t2 = 0; t2 < 2; ++t2)
-#pragma unroll
-    for (int s = 0; s < 4; ++s)
-      kf_off[t2][s] = swz(t2 * 32 + q32, (s * 16 + 8 * h32) * 2);
+  // K fragment offsets on the fly (same bit trick as AF_OFF in dkv)
+  const int kf_row[2] = {q32 * 128, (32 + q32) * 128};
+  const int kf_xk = (q32 & 7) << 4;
+#define KF_OFF(t2, s) (kf_row[t2] + (((s) * 32 + 16 * h32) ^ kf_xk))
   int trb[2][2];
   tr_bases(lane, trb);
   const float qscale = scale * LOG2E;
@@ -277,12 +275,19 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
   const int row_lo = qb * BM + w * 32;
   const int row_me = row_lo + q32;
   const int n_kv = (qb + 1) * BM / KVB;
+  short8v pend_k[Stage<NT>::REPS], pend_v[Stage<NT>::REPS];
+  stage_k.fetch(pend_k);
+  stage_v.fetch(pend_v);
   for (int j = 0; j < n_kv; ++j) {
     __syncthreads();
-    stage_k.run(lds_k);
-    stage_v.run(lds_v);
-    stage_k.advance();
-    stage_v.advance();
+    stage_k.put(lds_k, pend_k);
+    stage_v.put(lds_v, pend_v);
+    if (j + 1 < n_kv) {
+      stage_k.advance();
+      stage_v.advance();
+      stage_k.fetch(pend_k);
+      stage_v.fetch(pend_v);
+    }
     __syncthreads();
 
     const int key0 = j * KVB;
@@ -296,7 +301,7 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
       f32x16 acc = {};
 #pragma unroll
       for (int s = 0; s < 4; ++s)
-        acc = MFMA32(lds_read16(lds_k, kf_off[t2][s]), q_frag[s], acc);
+        acc = MFMA32(lds_read16(lds_k, KF_OFF(t2, s)), q_frag[s], acc);
       st[t2] = acc;
     }
     __builtin_amdgcn_s_setprio(0);
