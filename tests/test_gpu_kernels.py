@@ -332,3 +332,24 @@ def test_layernorm_fused_bwd_gpu():
     ref = torch.nn.functional.layer_norm(hf, (N,), wf, bf)
     ref.backward(dy.float())
     _close(dx, hf.grad + dh.float(), 8e-2, "ln fused dx+dh")
+
+
+def test_attention_rescale_spike_gpu():
+    """Force the online-softmax rescale mid-sequence (rule-of-thumb from the
+    CDNA guide: a rare data-dependent branch needs its own test): one K row
+    deep in the sequence dot-products hugely with every query, so the
+    running max jumps at a late tile and every O accumulator must rescale."""
+    torch.manual_seed(5)
+    B, H, T, D = 1, 2, 512, 64
+    q = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16)
+    k[:, :, 300, :] = 4.0 * torch.sign(q.mean(dim=2))  # spike vs all queries
+    scale = 1.0 / math.sqrt(D)
+    o, lse = _ext.get_ext().attention_fwd(q, k, v, scale)
+    ref = _attn_ref(q, k, v, scale)
+    _close(o, ref, 2e-2, "attn spike fwd")
+    s = torch.matmul(q.float(), k.float().transpose(-2, -1)) * scale
+    mask = torch.ones(T, T, dtype=torch.bool, device=q.device).tril()
+    lse_ref = torch.logsumexp(s.masked_fill(~mask, float("-inf")), dim=-1)
+    _close(lse, lse_ref, 2e-2, "attn spike lse")
