@@ -42,6 +42,9 @@ hipError_t tdsa_adamw_multi(const void*, long long, int, int, float, float,
                             float, float, float, long long, hipStream_t);
 int tdsa_adamw_desc_size();
 int tdsa_adamw_chunkref_size();
+int tdsa_sgd_desc_size();
+hipError_t tdsa_sgd_multi(const void*, long long, int, int, float, float,
+                          float, float, int, int, int, hipStream_t);
 hipError_t tdsa_sgd_step(void*, const void*, float*, float*, int, int, float,
                          float, float, float, int, int, int, long long, int,
                          int, hipStream_t);
@@ -320,6 +323,52 @@ void adamw_step_multi(std::vector<at::Tensor> params,
             "adamw_step_multi");
 }
 
+void sgd_step_multi(std::vector<at::Tensor> params,
+                    std::vector<at::Tensor> grads,
+                    std::vector<c10::optional<at::Tensor>> bufs,
+                    std::vector<c10::optional<at::Tensor>> masters,
+                    double lr, double momentum, double dampening, double wd,
+                    bool nesterov, bool maximize, bool first_step) {
+  const size_t n = params.size();
+  TORCH_CHECK(grads.size() == n && bufs.size() == n && masters.size() == n,
+              "length mismatch");
+  if (n == 0) return;
+  struct Desc {  // must mirror SgdTensorDesc in optim.hip
+    void* p; const void* g; float* buf; float* master;
+    long long numel; int param_bf16; int grad_bf16;
+  };
+  struct CRef { int tensor; int chunk; };
+  TORCH_CHECK((int)sizeof(Desc) == tdsa_sgd_desc_size());
+  const int chunk_elems = 1 << 16;
+  std::vector<Desc> descs(n);
+  std::vector<CRef> chunks;
+  for (size_t i = 0; i < n; ++i) {
+    Desc d;
+    d.p = params[i].data_ptr();
+    d.g = grads[i].data_ptr();
+    d.buf = bufs[i].has_value() ? bufs[i]->data_ptr<float>() : nullptr;
+    d.master = masters[i].has_value() ? masters[i]->data_ptr<float>() : nullptr;
+    d.numel = params[i].numel();
+    d.param_bf16 = dtype_flag(params[i]);
+    d.grad_bf16 = dtype_flag(grads[i]);
+    descs[i] = d;
+    const long long nch = (d.numel + chunk_elems - 1) / chunk_elems;
+    for (long long c = 0; c < nch; ++c) chunks.push_back({(int)i, (int)c});
+  }
+  const long long desc_bytes = (long long)(n * sizeof(Desc));
+  const long long total = desc_bytes + (long long)(chunks.size() * sizeof(CRef));
+  auto host = at::empty({total}, at::TensorOptions().dtype(at::kByte));
+  std::memcpy(host.data_ptr(), descs.data(), desc_bytes);
+  std::memcpy((char*)host.data_ptr() + desc_bytes, chunks.data(),
+              chunks.size() * sizeof(CRef));
+  auto dev = host.to(params[0].device());
+  check_hip(tdsa_sgd_multi(dev.data_ptr(), desc_bytes, (int)chunks.size(),
+                           chunk_elems, (float)lr, (float)momentum,
+                           (float)dampening, (float)wd, nesterov, maximize,
+                           first_step, cur_stream()),
+            "sgd_step_multi");
+}
+
 void sgd_step(at::Tensor param, at::Tensor grad, at::Tensor buf,
               at::Tensor master, bool has_buf, bool has_master, double lr,
               double momentum, double dampening, double wd, bool nesterov,
@@ -452,6 +501,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("adamw_step", &adamw_step);
   mod.def("adamw_step_multi", &adamw_step_multi);
   mod.def("sgd_step", &sgd_step);
+  mod.def("sgd_step_multi", &sgd_step_multi);
   mod.def("attention_fwd", &attention_fwd, py::arg("q"), py::arg("k"),
           py::arg("v"), py::arg("scale"), py::arg("out") = py::none());
   mod.def("attention_bwd", &attention_bwd, py::arg("q"), py::arg("k"),

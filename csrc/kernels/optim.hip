@@ -123,9 +123,85 @@ __global__ void adamw_multi_kernel(const AdamTensorDesc* __restrict__ descs,
   }
 }
 
+template <typename PT, typename GT>
+DEV_INLINE void sgd_update_span(PT* p, const GT* g, float* buf, float* master,
+                                long long start, long long end, float lr,
+                                float momentum, float dampening, float wd,
+                                int nesterov, int maximize, int first_step) {
+  for (long long i = start + threadIdx.x; i < end; i += blockDim.x) {
+    float gv = (float)g[i];
+    if (maximize) gv = -gv;
+    float pv = master ? master[i] : (float)p[i];
+    if (wd != 0.0f) gv += wd * pv;
+    if (buf) {
+      float b = first_step ? gv : buf[i] * momentum + (1.0f - dampening) * gv;
+      buf[i] = b;
+      gv = nesterov ? (gv + momentum * b) : b;
+    }
+    pv -= lr * gv;
+    if (master) master[i] = pv;
+    p[i] = (PT)pv;
+  }
+}
+
+struct SgdTensorDesc {
+  void* p;
+  const void* g;
+  float* buf;        // nullptr without momentum
+  float* master;     // nullptr when param is fp32
+  long long numel;
+  int param_bf16;
+  int grad_bf16;
+};
+
+__global__ void sgd_multi_kernel(const SgdTensorDesc* __restrict__ descs,
+                                 const ChunkRef* __restrict__ chunks,
+                                 int nchunks, int chunk_elems, float lr,
+                                 float momentum, float dampening, float wd,
+                                 int nesterov, int maximize, int first_step) {
+  for (int cid = blockIdx.x; cid < nchunks; cid += gridDim.x) {
+    const ChunkRef ch = chunks[cid];
+    const SgdTensorDesc d = descs[ch.tensor];
+    const long long start = (long long)ch.chunk * chunk_elems;
+    const long long end = min(d.numel, start + chunk_elems);
+    if (d.param_bf16 && d.grad_bf16)
+      sgd_update_span((bf16*)d.p, (const bf16*)d.g, d.buf, d.master, start,
+                      end, lr, momentum, dampening, wd, nesterov, maximize,
+                      first_step);
+    else if (d.param_bf16)
+      sgd_update_span((bf16*)d.p, (const float*)d.g, d.buf, d.master, start,
+                      end, lr, momentum, dampening, wd, nesterov, maximize,
+                      first_step);
+    else if (d.grad_bf16)
+      sgd_update_span((float*)d.p, (const bf16*)d.g, d.buf, d.master, start,
+                      end, lr, momentum, dampening, wd, nesterov, maximize,
+                      first_step);
+    else
+      sgd_update_span((float*)d.p, (const float*)d.g, d.buf, d.master, start,
+                      end, lr, momentum, dampening, wd, nesterov, maximize,
+                      first_step);
+  }
+}
+
 }  // namespace
 
 extern "C" {
+
+int tdsa_sgd_desc_size() { return (int)sizeof(SgdTensorDesc); }
+
+hipError_t tdsa_sgd_multi(const void* blob, long long desc_bytes, int nchunks,
+                          int chunk_elems, float lr, float momentum,
+                          float dampening, float wd, int nesterov, int maximize,
+                          int first_step, hipStream_t stream) {
+  const SgdTensorDesc* descs = (const SgdTensorDesc*)blob;
+  const ChunkRef* chunks = (const ChunkRef*)((const char*)blob + desc_bytes);
+  int grid = nchunks < 2048 ? nchunks : 2048;
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(sgd_multi_kernel, dim3(grid), dim3(256), 0, stream, descs,
+                     chunks, nchunks, chunk_elems, lr, momentum, dampening, wd,
+                     nesterov, maximize, first_step);
+  return hipGetLastError();
+}
 
 // blob layout: [ndescs x AdamTensorDesc][nchunks x ChunkRef], device memory.
 hipError_t tdsa_adamw_multi(const void* blob, long long desc_bytes, int nchunks,

@@ -353,3 +353,27 @@ def test_attention_rescale_spike_gpu():
     mask = torch.ones(T, T, dtype=torch.bool, device=q.device).tril()
     lse_ref = torch.logsumexp(s.masked_fill(~mask, float("-inf")), dim=-1)
     _close(lse, lse_ref, 2e-2, "attn spike lse")
+
+
+def test_sgd_multi_matches_single_gpu():
+    from tiny_deepspeed_amd import SGD
+
+    torch.manual_seed(0)
+    shapes = [(100, 32), (7,), (513,)]
+    p1 = [torch.nn.Parameter(torch.randn(s, device="cuda", dtype=torch.bfloat16))
+          for s in shapes]
+    p2 = [torch.nn.Parameter(t.detach().clone()) for t in p1]
+    o1 = SGD([(f"p{i}", p) for i, p in enumerate(p1)], lr=0.1, momentum=0.9,
+             weight_decay=0.01)
+    o2 = SGD([(f"p{i}", p) for i, p in enumerate(p2)], lr=0.1, momentum=0.9,
+             weight_decay=0.01)
+    o2._apply_updates = lambda items: [o2.one_step(n, p) for n, p in items]
+    for it in range(3):
+        for a, b in zip(p1, p2):
+            g = torch.randn_like(a)
+            a.grad = g
+            b.grad = g.clone()
+        o1.step()
+        o2.step()
+    for a, b in zip(p1, p2):
+        _close(a.data, b.data, 1e-6, "sgd multi vs single")

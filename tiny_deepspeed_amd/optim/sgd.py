@@ -42,6 +42,27 @@ class SGD(Optimizer):
             self.master[name] = p.detach().float().clone()
 
     @torch.no_grad()
+    def _apply_updates(self, items):
+        if not items or not items[0][1].is_cuda or not ops.ext_available():
+            return super()._apply_updates(items)
+        params, grads, bufs, masters = [], [], [], []
+        first = all(n not in self._stepped for n, _ in items)
+        mixed = any((n in self._stepped) != (not first) for n, _ in items)
+        if mixed:  # per-tensor first-step flags differ: fall back
+            return super()._apply_updates(items)
+        for name, p in items:
+            params.append(p.data)
+            grads.append(p.grad)
+            bufs.append(self.velocities.get(name))
+            masters.append(self.master.get(name))
+            self._stepped.add(name)
+        ops.get_ext().sgd_step_multi(
+            params, grads, bufs, masters, self.lr, self.momentum,
+            self.dampening, self.weight_decay, self.nesterov, self.maximize,
+            first,
+        )
+
+    @torch.no_grad()
     def one_step(self, name, param):
         first = name not in self._stepped
         self._stepped.add(name)
