@@ -89,7 +89,7 @@ def main():
         # shrink so it completes in seconds.
         args.model = "gpt2-small"
         args.batch = 1
-        args.seq = min(args.seq, 128)
+        args.seq = min(args.seq, 64)
 
     model_name = args.model
     config = GPTConfig.named(model_name, block_size=max(args.seq, 64))
