@@ -54,6 +54,11 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
   const int lane = tid & (WAVE - 1);
   const int wid = tid / WAVE;
   const int nw = blockDim.x / WAVE;
+  // block-local loss/count accumulators: one atomicAdd per BLOCK at the end
+  // (a per-row atomicAdd on the single loss scalar serialized 32k blocks —
+  // measured 4x the kernel's HBM floor)
+  float loss_acc = 0.f;
+  int valid_acc = 0;
   for (long long row = blockIdx.x; row < R; row += gridDim.x) {
     const T* lr = logits + row * V;
     float m = -INFINITY, s = 0.f;
@@ -106,12 +111,16 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
         const long long t = targets[row];
         if (t != ignore_index) {
           float picked = (float)lr[t];
-          atomicAdd(loss_sum, l - picked);
-          atomicAdd(n_valid, 1);
+          loss_acc += l - picked;
+          valid_acc += 1;
         }
       }
     }
     __syncthreads();
+  }
+  if (tid == 0 && valid_acc > 0) {
+    atomicAdd(loss_sum, loss_acc);
+    atomicAdd(n_valid, valid_acc);
   }
 }
 
