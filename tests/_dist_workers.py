@@ -180,3 +180,31 @@ def grad_accumulation(rank, world):
         model(x).square().mean().backward()
     model.comm.sync()
     return lin.weight.grad.clone()
+
+
+def broadcast_bucketed_roundtrip(rank, world):
+    """Mixed owners/sizes through comm.broadcast_bucketed: every rank ends
+    with the owner's values (small tensors ride flat buckets)."""
+    from tiny_deepspeed_amd.parallel.comm import CommContext
+
+    comm = CommContext()
+    torch.manual_seed(100 + rank)  # ranks start DIFFERENT
+    sizes = [(3,), (100,), (5, 5), (1 << 20,), (7,), (2, 2)]
+    owners = [0, 1, 1, 0, 1, 0]
+    tensors = [torch.randn(s) for s in sizes]
+    expected = []
+    for t, owner in zip(tensors, owners):
+        g = torch.Generator().manual_seed(100 + owner)
+        # regenerate what the owner drew for THIS tensor: replay its stream
+        expected.append(None)  # filled below
+    # replay each owner's full stream to know its values
+    for owner in (0, 1):
+        g = torch.random.manual_seed(100 + owner)
+        vals = [torch.randn(s) for s in sizes]
+        for i, o in enumerate(owners):
+            if o == owner:
+                expected[i] = vals[i]
+    comm.broadcast_bucketed([(t, o) for t, o in zip(tensors, owners)],
+                            bucket_bytes=1024)
+    comm.sync()
+    return all(torch.equal(t, e) for t, e in zip(tensors, expected))

@@ -75,3 +75,9 @@ def test_grad_accumulation_syncs_total():
             lin(x).square().mean().backward()
         total += lin.weight.grad
     assert torch.allclose(results[0], total / 2, atol=1e-6)
+
+
+def test_broadcast_bucketed_groups_small_tensors():
+    results = run_distributed(W.broadcast_bucketed_roundtrip, world=2)
+    for rank, ok in results.items():
+        assert ok, f"rank {rank}"
