@@ -1,4 +1,4 @@
-"""ZeRO-1 GPT-2 training (parity: /root/reference/example/zero2/train.py).
+"""ZeRO-2 GPT-2 training (parity: /root/reference/example/zero2/train.py).
 
 Partition is planned on the meta device (no allocation), then the model is
 materialized and wrapped.

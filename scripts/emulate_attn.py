@@ -1,4 +1,7 @@
 """CPU emulation of attention.hip's lane-level dataflow (fwd, one workgroup)
+NOTE: this script models the RETIRED 16x16x32 kernel variant (the shipped
+kernels use 32x32x16); kept as the lane-level debugging method record that
+located the cross-lane repack bug (see docs/kernels.md).
 to locate logic bugs without GPU round-trips.
 
 MFMA semantics (validated on hardware by scripts/debug_mfma.py):
