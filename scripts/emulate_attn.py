@@ -1,8 +1,9 @@
-"""CPU emulation of attention.hip's lane-level dataflow (fwd, one workgroup)
-NOTE: this script models the RETIRED 16x16x32 kernel variant (the shipped
-kernels use 32x32x16); kept as the lane-level debugging method record that
-located the cross-lane repack bug (see docs/kernels.md).
-to locate logic bugs without GPU round-trips.
+"""CPU emulation of the attention forward's lane-level dataflow (one
+workgroup) used to locate logic bugs without GPU round-trips.
+
+NOTE: models the RETIRED 16x16x32 kernel variant (the shipped kernels use
+32x32x16); kept as the record of the debugging method that located the
+cross-lane P-repack bug (see docs/kernels.md).
 
 MFMA semantics (validated on hardware by scripts/debug_mfma.py):
   D[i][j] = sum over g in 0..3, e in 0..7 of
