@@ -56,6 +56,7 @@ hipError_t tdsa_attn_bwd(const void*, const void*, const void*, const void*,
                          long long, long long, int, float, const long long*,
                          const long long*, const long long*, hipStream_t);
 hipError_t tdsa_dbg_mfma(const void*, const void*, float*, int, hipStream_t);
+hipError_t tdsa_dbg_mfma32(const void*, const void*, float*, hipStream_t);
 hipError_t tdsa_dbg_stage(const void*, void*, int, hipStream_t);
 hipError_t tdsa_dbg_tr16(const void*, float*, hipStream_t);
 }
@@ -462,6 +463,15 @@ at::Tensor dbg_mfma(at::Tensor A, at::Tensor B, int64_t variant) {
   return D;
 }
 
+at::Tensor dbg_mfma32(at::Tensor A, at::Tensor B) {
+  CHECK_IN(A); CHECK_IN(B);
+  auto D = at::zeros({32, 32}, A.options().dtype(at::kFloat));
+  check_hip(tdsa_dbg_mfma32(A.data_ptr(), B.data_ptr(), D.data_ptr<float>(),
+                            cur_stream()),
+            "dbg_mfma32");
+  return D;
+}
+
 at::Tensor dbg_stage(at::Tensor in, int64_t transposed) {
   CHECK_IN(in);
   auto out = at::zeros_like(in);
@@ -483,6 +493,7 @@ at::Tensor dbg_tr16(at::Tensor in) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("dbg_mfma", &dbg_mfma);
+  mod.def("dbg_mfma32", &dbg_mfma32);
   mod.def("dbg_stage", &dbg_stage);
   mod.def("dbg_tr16", &dbg_tr16);
   mod.def("layernorm_fwd", &layernorm_fwd, py::arg("x"), py::arg("w"),

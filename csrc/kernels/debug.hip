@@ -37,6 +37,28 @@ __global__ void dbg_mfma_kernel(const bf16* __restrict__ A,
   for (int r = 0; r < 4; ++r) D[(4 * g + r) * 16 + c] = acc[r];
 }
 
+// 32x32x16 probe with the PRODUCTION layouts (attention.hip):
+// A[l%32][(l/32)*8+e], B[(l/32)*8+e][l%32], C[(r&3)+8*(r>>2)+4*(l/32)][l%32]
+__global__ void dbg_mfma32_kernel(const bf16* __restrict__ A,
+                                  const bf16* __restrict__ B,
+                                  float* __restrict__ D) {
+  const int lane = threadIdx.x;
+  const int m32 = lane & 31;
+  const int h = lane >> 5;
+  union { bfrag f; short s[8]; } a, b;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    const int k = 8 * h + e;
+    a.s[e] = *reinterpret_cast<const short*>(A + m32 * 16 + k);
+    b.s[e] = *reinterpret_cast<const short*>(B + k * 32 + m32);
+  }
+  f32x16 acc = {};
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a.f, b.f, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 16; ++r)
+    D[((r & 3) + 8 * (r >> 2) + 4 * h) * 32 + m32] = acc[r];
+}
+
 __global__ void dbg_stage_kernel(const bf16* __restrict__ in,
                                  bf16* __restrict__ out, int transposed) {
   __shared__ __attribute__((aligned(16))) char lds[64 * 64 * 2];
@@ -111,6 +133,13 @@ extern "C" {
 hipError_t tdsa_dbg_tr16(const void* in, float* out, hipStream_t stream) {
   hipLaunchKernelGGL(dbg_tr16_kernel, dim3(1), dim3(64), 0, stream,
                      (const bf16*)in, out);
+  return hipGetLastError();
+}
+
+hipError_t tdsa_dbg_mfma32(const void* A, const void* B, float* D,
+                           hipStream_t stream) {
+  hipLaunchKernelGGL(dbg_mfma32_kernel, dim3(1), dim3(64), 0, stream,
+                     (const bf16*)A, (const bf16*)B, D);
   return hipGetLastError();
 }
 

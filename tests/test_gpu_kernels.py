@@ -377,3 +377,36 @@ def test_sgd_multi_matches_single_gpu():
         o2.step()
     for a, b in zip(p1, p2):
         _close(a.data, b.data, 1e-6, "sgd multi vs single")
+
+
+def test_mfma_layout_probes_gpu():
+    """Regression-guard the hardware fragment-layout assumptions the
+    attention kernels rely on (asymmetric operands: transpose-detecting)."""
+    torch.manual_seed(0)
+    # 16x16x32 slot-pairing probe
+    A = torch.randint(-8, 8, (16, 32), device="cuda").to(torch.bfloat16)
+    B = torch.randint(-8, 8, (32, 16), device="cuda").to(torch.bfloat16)
+    D = _ext.get_ext().dbg_mfma(A, B, 0)
+    assert torch.equal(D, A.float() @ B.float())
+    # production 32x32x16 layouts
+    A2 = torch.randint(-8, 8, (32, 16), device="cuda").to(torch.bfloat16)
+    B2 = torch.randint(-8, 8, (16, 32), device="cuda").to(torch.bfloat16)
+    D2 = _ext.get_ext().dbg_mfma32(A2, B2)
+    assert torch.equal(D2, A2.float() @ B2.float())
+    # ds_read_tr16_b64 lane mapping (row / column identification)
+    Mr = torch.arange(64, device="cuda").view(64, 1).expand(64, 64) \
+        .contiguous().to(torch.bfloat16)
+    Mc = torch.arange(64, device="cuda").view(1, 64).expand(64, 64) \
+        .contiguous().to(torch.bfloat16)
+    Rr = _ext.get_ext().dbg_tr16(Mr)
+    Rc = _ext.get_ext().dbg_tr16(Mc)
+    l = torch.arange(64, device="cuda")
+    g4 = l >> 4
+    for dt in range(2):
+        for s in range(4):
+            for half in range(2):
+                for j in range(4):
+                    er = 16 * s + 8 * (g4 >> 1) + 4 * half + j
+                    ec = dt * 32 + 16 * (g4 & 1) + (l & 15)
+                    assert torch.equal(Rr[dt, s, half, j].long(), er), (dt, s)
+                    assert torch.equal(Rc[dt, s, half, j].long(), ec), (dt, s)
