@@ -238,7 +238,9 @@ hipError_t tdsa_ln_fwd(const void* x, const void* res, void* h, const void* w,
                        int M, int N, float eps, int is_bf16,
                        hipStream_t stream) {
   const int block = ln_block(N, is_bf16 ? 8 : 4);
-  const int grid = (M < 8192) ? M : 8192;
+  // one row per block up to 32k: chaining rows serializes the per-row
+  // reduction barriers; resident blocks overlap freely instead
+  const int grid = (M < 32768) ? M : 32768;
 #define LN_FWD(T, HASR)                                                       \
   hipLaunchKernelGGL((ln_fwd_kernel<T, HASR>), dim3(grid), dim3(block), 0,    \
                      stream, (const T*)x, (const T*)res, (T*)h, (const T*)w,  \
@@ -257,7 +259,7 @@ hipError_t tdsa_ln_fwd(const void* x, const void* res, void* h, const void* w,
 // G (stripe count) is chosen here and reported to the caller so it can size
 // pdw/pdb; call with pdw==nullptr to query G.
 int tdsa_ln_bwd_dx_stripes(int M) {
-  int g = M < 2048 ? M : 2048;
+  int g = M < 4096 ? M : 4096;
   return g < 1 ? 1 : g;
 }
 
