@@ -36,6 +36,8 @@
 // composite rocBLAS path otherwise, ops/attention.py).
 #include "common.h"
 
+#include <cstdlib>
+
 #define MFMA32(a, b, c) \
   __builtin_amdgcn_mfma_f32_32x32x16_bf16((a), (b), (c), 0, 0, 0)
 
@@ -721,6 +723,15 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
 
 extern "C" {
 
+// Within-box tuning knobs (defaults are the measured-best dispatch; the env
+// override exists for A/B because box-to-box wall-clock noise is ~3%).
+static int env_nw(const char* name, int dflt) {
+  const char* v = getenv(name);
+  if (!v) return dflt;
+  int n = atoi(v);
+  return (n == 2 || n == 4 || n == 8) ? n : dflt;
+}
+
 // strides arrays: {b, h, t} in elements, per tensor group:
 // sq = q/k/v, so = o (and dO in bwd), sd = dq/dk/dv.
 hipError_t tdsa_attn_fwd(const void* q, const void* k, const void* v, void* o,
@@ -735,8 +746,9 @@ hipError_t tdsa_attn_fwd(const void* q, const void* k, const void* v, void* o,
                      dim3(NW * WAVE), 0, stream, (const bf16*)q,              \
                      (const bf16*)k, (const bf16*)v, (bf16*)o, lse, T,        \
                      (int)H, scale, sq, so)
-  if (T % 256 == 0) LAUNCH_FWD(8);
-  else if (T % 128 == 0) LAUNCH_FWD(4);
+  const int nw_fwd = env_nw("TDSA_ATTN_FWD_NW", 8);
+  if (T % 256 == 0 && nw_fwd == 8) LAUNCH_FWD(8);
+  else if (T % 128 == 0 && nw_fwd >= 4) LAUNCH_FWD(4);
   else LAUNCH_FWD(2);
 #undef LAUNCH_FWD
   return hipGetLastError();
@@ -772,10 +784,11 @@ hipError_t tdsa_attn_bwd(const void* q, const void* k, const void* v,
                        (const bf16*)v, (const bf16*)dout, lse, delta,         \
                        (bf16*)dq, T, (int)H, scale, sq, so, sd);              \
   } while (0)
-  // 8-wave blocks measured 285us vs 332us for 4-wave at B8/H16/T1024:
-  // staging amortization across 8 waves beats the extra block-level overlap.
-  if (T % 256 == 0) LAUNCH_BWD(8);
-  else if (T % 128 == 0) LAUNCH_BWD(4);
+  // 8-wave blocks measured 285us vs 332us for 4-wave at B8/H16/T1024
+  // (pre-tr16); staging amortization across 8 waves beat block overlap.
+  const int nw_bwd = env_nw("TDSA_ATTN_BWD_NW", 8);
+  if (T % 256 == 0 && nw_bwd == 8) LAUNCH_BWD(8);
+  else if (T % 128 == 0 && nw_bwd >= 4) LAUNCH_BWD(4);
   else LAUNCH_BWD(2);
 #undef LAUNCH_BWD
   return hipGetLastError();
