@@ -9,6 +9,8 @@
 // (SURVEY.md 2.10A).
 #include "common.h"
 
+#include <cstdlib>
+
 namespace {
 
 template <typename T> struct VecTraits;
@@ -238,9 +240,11 @@ hipError_t tdsa_ln_fwd(const void* x, const void* res, void* h, const void* w,
                        int M, int N, float eps, int is_bf16,
                        hipStream_t stream) {
   const int block = ln_block(N, is_bf16 ? 8 : 4);
-  // one row per block up to 32k: chaining rows serializes the per-row
+  // one row per block up to the cap: chaining rows serializes the per-row
   // reduction barriers; resident blocks overlap freely instead
-  const int grid = (M < 32768) ? M : 32768;
+  int cap = 32768;
+  if (const char* v = getenv("TDSA_LN_GRID")) cap = atoi(v);
+  const int grid = (M < cap) ? M : cap;
 #define LN_FWD(T, HASR)                                                       \
   hipLaunchKernelGGL((ln_fwd_kernel<T, HASR>), dim3(grid), dim3(block), 0,    \
                      stream, (const T*)x, (const T*)res, (T*)h, (const T*)w,  \
@@ -259,7 +263,10 @@ hipError_t tdsa_ln_fwd(const void* x, const void* res, void* h, const void* w,
 // G (stripe count) is chosen here and reported to the caller so it can size
 // pdw/pdb; call with pdw==nullptr to query G.
 int tdsa_ln_bwd_dx_stripes(int M) {
-  int g = M < 4096 ? M : 4096;
+  // within-box sweep: 2048 stripes beat 4096 (dwdb reduce pays per stripe)
+  int cap = 2048;
+  if (const char* v = getenv("TDSA_LN_STRIPES")) cap = atoi(v);
+  int g = M < cap ? M : cap;
   return g < 1 ? 1 : g;
 }
 
