@@ -7,6 +7,8 @@
 // pass writing (softmax - onehot) * scale. fp32 accumulation throughout.
 #include "common.h"
 
+#include <cstdlib>
+
 namespace {
 
 // merge two (m, s) logsumexp states (raw v_exp: exp(-inf) == 0, and the
@@ -174,7 +176,9 @@ hipError_t tdsa_ce_fwd(const void* logits, const long long* targets, float* lse,
                        float* loss_sum, int* n_valid, long long R, int V,
                        long long ignore_index, int is_bf16, hipStream_t stream) {
   const int block = 256;
-  const int grid = (int)((R < 4096) ? R : 4096);
+  long long cap = 32768;
+  if (const char* v = getenv("TDSA_CE_GRID")) cap = atoll(v);
+  const int grid = (int)((R < cap) ? R : cap);
   if (is_bf16)
     hipLaunchKernelGGL((ce_fwd_kernel<bf16, 8>), dim3(grid), dim3(block), 0,
                        stream, (const bf16*)logits, targets, lse, loss_sum,
@@ -191,7 +195,9 @@ hipError_t tdsa_ce_bwd(const void* logits, const long long* targets,
                        float scale, long long ignore_index, int is_bf16,
                        hipStream_t stream) {
   const int block = 256;
-  const int grid = (int)((R < 4096) ? R : 4096);
+  long long cap = 32768;
+  if (const char* v = getenv("TDSA_CE_GRID")) cap = atoll(v);
+  const int grid = (int)((R < cap) ? R : cap);
   if (is_bf16)
     hipLaunchKernelGGL((ce_bwd_kernel<bf16, 8>), dim3(grid), dim3(block), 0,
                        stream, (const bf16*)logits, targets, lse, (bf16*)dlogits,
