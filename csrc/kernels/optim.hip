@@ -82,20 +82,37 @@ struct ChunkRef {
 };
 
 template <typename PT, typename GT>
+DEV_INLINE void adamw_elem(PT* p, const GT* g, float* m, float* v,
+                           float* master, long long i, float lr, float b1,
+                           float b2, float eps, float wd, float inv_bc1,
+                           float inv_bc2) {
+  float gv = (float)g[i];
+  float pv = master ? master[i] : (float)p[i];
+  pv *= (1.0f - lr * wd);
+  float mi = m[i] = m[i] * b1 + (1.0f - b1) * gv;
+  float vi = v[i] = v[i] * b2 + (1.0f - b2) * gv * gv;
+  pv -= lr * inv_bc1 * mi / (sqrtf(vi * inv_bc2) + eps);
+  if (master) master[i] = pv;
+  p[i] = (PT)pv;
+}
+
+// 4 elements per lane: the optimizer state (m, v, master) is fp32 and
+// dominates the traffic; 16B-per-lane accesses beat scalar 4B.
+template <typename PT, typename GT>
 DEV_INLINE void adamw_update_span(PT* p, const GT* g, float* m, float* v,
                                   float* master, long long start, long long end,
                                   float lr, float b1, float b2, float eps,
                                   float wd, float inv_bc1, float inv_bc2) {
-  for (long long i = start + threadIdx.x; i < end; i += blockDim.x) {
-    float gv = (float)g[i];
-    float pv = master ? master[i] : (float)p[i];
-    pv *= (1.0f - lr * wd);
-    float mi = m[i] = m[i] * b1 + (1.0f - b1) * gv;
-    float vi = v[i] = v[i] * b2 + (1.0f - b2) * gv * gv;
-    pv -= lr * inv_bc1 * mi / (sqrtf(vi * inv_bc2) + eps);
-    if (master) master[i] = pv;
-    p[i] = (PT)pv;
+  const long long n4 = (end - start) / 4;
+  for (long long q = threadIdx.x; q < n4; q += blockDim.x) {
+    const long long i = start + q * 4;
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+      adamw_elem(p, g, m, v, master, i + k, lr, b1, b2, eps, wd, inv_bc1,
+                 inv_bc2);
   }
+  for (long long i = start + n4 * 4 + threadIdx.x; i < end; i += blockDim.x)
+    adamw_elem(p, g, m, v, master, i, lr, b1, b2, eps, wd, inv_bc1, inv_bc2);
 }
 
 __global__ void adamw_multi_kernel(const AdamTensorDesc* __restrict__ descs,
