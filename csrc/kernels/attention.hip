@@ -270,6 +270,7 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
     q_frag[s] = load_frag_scaled(qp, w * 32 + q32, s * 16 + 8 * h32, sq.t,
                                  qscale);
 
+  const f32x16 kzero = {};
   f32x16 o_acc[2] = {};
   float m_run = -INFINITY;
   float l_run = 0.f;
@@ -300,9 +301,11 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
     __builtin_amdgcn_s_setprio(1);  // favor MFMA-issuing waves (T5)
 #pragma unroll
     for (int t2 = 0; t2 < 2; ++t2) {
-      f32x16 acc = {};
+      // seed from the loop-invariant zero vector: a per-tile `= {}` init
+      // re-emits 16 v_mov per accumulator per tile
+      f32x16 acc = MFMA32(lds_read16(lds_k, KF_OFF(t2, 0)), q_frag[0], kzero);
 #pragma unroll
-      for (int s = 0; s < 4; ++s)
+      for (int s = 1; s < 4; ++s)
         acc = MFMA32(lds_read16(lds_k, KF_OFF(t2, s)), q_frag[s], acc);
       st[t2] = acc;
     }
@@ -484,6 +487,7 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
     v_frag[s] = load_frag(vp, key_lo + k32, s * 16 + 8 * h32, sq.t);
   }
 
+  const f32x16 kzero = {};
   f32x16 dk_acc[2] = {};
   f32x16 dv_acc[2] = {};
 
@@ -515,11 +519,11 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
 #pragma unroll
     for (int t2 = 0; t2 < 2; ++t2) {
       // S, dP tiles: C[qrow = 32*t2 + crow(r,h32)][key = k32]
-      f32x16 s_acc = {};
-      f32x16 dp_acc = {};
       __builtin_amdgcn_s_setprio(1);
+      f32x16 s_acc = MFMA32(lds_read16(lds_q, AF_OFF(t2, 0)), k_frag[0], kzero);
+      f32x16 dp_acc = MFMA32(lds_read16(lds_do, AF_OFF(t2, 0)), v_frag[0], kzero);
 #pragma unroll
-      for (int s = 0; s < 4; ++s) {
+      for (int s = 1; s < 4; ++s) {
         s_acc = MFMA32(lds_read16(lds_q, AF_OFF(t2, s)), k_frag[s], s_acc);
         dp_acc = MFMA32(lds_read16(lds_do, AF_OFF(t2, s)), v_frag[s], dp_acc);
       }
@@ -640,6 +644,7 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   const float lse2_me = lse[bh * T + row_me] * LOG2E;
   const float dlt_me = delta[bh * T + row_me];
 
+  const f32x16 kzero = {};
   f32x16 dq_acc[2] = {};
 
   const int n_kv = (qb + 1) * BM / KVB;
@@ -666,11 +671,11 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
 #pragma unroll
     for (int t2 = 0; t2 < 2; ++t2) {
       // S^T, dP^T tiles: C[key = 32*t2 + crow(r,h32)][qrow = q32]
-      f32x16 s_acc = {};
-      f32x16 dp_acc = {};
       __builtin_amdgcn_s_setprio(1);
+      f32x16 s_acc = MFMA32(lds_read16(lds_k, f_off[t2][0]), q_frag[0], kzero);
+      f32x16 dp_acc = MFMA32(lds_read16(lds_v, f_off[t2][0]), do_frag[0], kzero);
 #pragma unroll
-      for (int s = 0; s < 4; ++s) {
+      for (int s = 1; s < 4; ++s) {
         s_acc = MFMA32(lds_read16(lds_k, f_off[t2][s]), q_frag[s], s_acc);
         dp_acc = MFMA32(lds_read16(lds_v, f_off[t2][s]), do_frag[s], dp_acc);
       }
