@@ -399,36 +399,6 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
 }
 
 // ---------------------------------------------------------------------------
-// Backward: delta = rowsum(dO * O)
-// ---------------------------------------------------------------------------
-// grid: (ceil(T/rows_per_block), B*H); 8 rows per wave, 8 bf16 per lane
-// (a one-element-per-lane version loaded 2 B per lane and ran at 1.5 TB/s).
-__global__ void attn_delta_kernel(const bf16* __restrict__ dout,
-                                  const bf16* __restrict__ o,
-                                  float* __restrict__ delta, int T, int H,
-                                  GStride so) {
-  constexpr int ROWS_PER_WAVE = 8;  // D=64 = 8 lanes x 8 elements
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int wid = threadIdx.x / WAVE;
-  const int t = (blockIdx.x * (blockDim.x / WAVE) + wid) * ROWS_PER_WAVE
-                + (lane >> 3);
-  if (t >= T) return;
-  const long long bh = blockIdx.y;
-  const long long off = (bh / H) * so.b + (bh % H) * so.h
-                        + (long long)t * so.t + (lane & 7) * 8;
-  short8v dv = load8(dout + off);
-  short8v ov = load8(o + off);
-  float acc = 0.f;
-#pragma unroll
-  for (int k = 0; k < 8; ++k) acc += bf_elem(dv, k) * bf_elem(ov, k);
-  // reduce across the 8 lanes sharing a row
-  acc += __shfl_xor(acc, 1, WAVE);
-  acc += __shfl_xor(acc, 2, WAVE);
-  acc += __shfl_xor(acc, 4, WAVE);
-  if ((lane & 7) == 0) delta[bh * T + t] = acc;
-}
-
-// ---------------------------------------------------------------------------
 // Backward dK/dV: one workgroup per (NW*32)-key block; wave w owns 32 keys.
 // ---------------------------------------------------------------------------
 template <int NW>
@@ -593,9 +563,10 @@ __launch_bounds__(NW * WAVE)  // capping at 128 VGPR spills 164 B/lane here
 __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
                                    const bf16* __restrict__ k,
                                    const bf16* __restrict__ v,
+                                   const bf16* __restrict__ o,
                                    const bf16* __restrict__ dout,
                                    const float* __restrict__ lse,
-                                   const float* __restrict__ delta,
+                                   float* __restrict__ delta,
                                    bf16* __restrict__ dq, int T, int H,
                                    float scale, GStride sq, GStride so,
                                    GStride sd) {
@@ -642,7 +613,23 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   const int row_lo = qb * BM + w * 32;
   const int row_me = row_lo + q32;
   const float lse2_me = lse[bh * T + row_me] * LOG2E;
-  const float dlt_me = delta[bh * T + row_me];
+  // delta = rowsum(dO*O) computed here from the fragments already in
+  // registers (+ one O load) and PUBLISHED for the dkv kernel, which is
+  // launched after this one — replaces a separate full-pass delta kernel.
+  float dlt_me = 0.f;
+#pragma unroll
+  for (int s = 0; s < 4; ++s) {
+    union { bfrag f; short8v v8; } ov;
+    ov.f = load_frag(o + ooff + (long long)(qb * BM) * so.t, w * 32 + q32,
+                     s * 16 + 8 * h32, so.t);
+    union { bfrag f; short8v v8; } dv;
+    dv.f = do_frag[s];
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      dlt_me += bf_elem(ov.v8, e) * bf_elem(dv.v8, e);
+  }
+  dlt_me += __shfl_xor(dlt_me, 32, WAVE);
+  if (lane < 32) delta[bh * T + row_me] = dlt_me;
 
   const f32x16 kzero = {};
   f32x16 dq_acc[2] = {};
@@ -770,24 +757,19 @@ hipError_t tdsa_attn_bwd(const void* q, const void* k, const void* v,
   GStride so{so_in[0], so_in[1], (int)so_in[2]};
   GStride sd{sd_in[0], sd_in[1], (int)sd_in[2]};
   const long long BH = B * H;
-  {
-    const int rows_per_block = (256 / WAVE) * 8;
-    hipLaunchKernelGGL(attn_delta_kernel,
-                       dim3((T + rows_per_block - 1) / rows_per_block, BH),
-                       dim3(256), 0, stream, (const bf16*)dout, (const bf16*)o,
-                       delta, T, (int)H, so);
-  }
+  // dq runs FIRST: it computes and publishes delta = rowsum(dO*O) from
+  // fragments it loads anyway; dkv (same stream) consumes it.
 #define LAUNCH_BWD(NW)                                                        \
   do {                                                                        \
     dim3 grid(T / (NW * 32), BH);                                             \
+    hipLaunchKernelGGL(attn_bwd_dq_kernel<NW>, grid, dim3(NW * WAVE), 0,      \
+                       stream, (const bf16*)q, (const bf16*)k,                \
+                       (const bf16*)v, (const bf16*)o, (const bf16*)dout,     \
+                       lse, delta, (bf16*)dq, T, (int)H, scale, sq, so, sd);  \
     hipLaunchKernelGGL(attn_bwd_dkv_kernel<NW>, grid, dim3(NW * WAVE), 0,     \
                        stream, (const bf16*)q, (const bf16*)k,                \
                        (const bf16*)v, (const bf16*)dout, lse, delta,         \
                        (bf16*)dk, (bf16*)dv, T, (int)H, scale, sq, so, sd);   \
-    hipLaunchKernelGGL(attn_bwd_dq_kernel<NW>, grid, dim3(NW * WAVE), 0,      \
-                       stream, (const bf16*)q, (const bf16*)k,                \
-                       (const bf16*)v, (const bf16*)dout, lse, delta,         \
-                       (bf16*)dq, T, (int)H, scale, sq, so, sd);              \
   } while (0)
   // 8-wave blocks measured 285us vs 332us for 4-wave at B8/H16/T1024
   // (pre-tr16); staging amortization across 8 waves beat block overlap.
