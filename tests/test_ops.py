@@ -146,3 +146,31 @@ def test_autotuner_picks_and_caches():
     tuner.choose("op", [fa, fb], x)
     # cached choice: exactly one more call total
     assert calls["a"] + calls["b"] == before["a"] + before["b"] + 1
+
+
+def test_fused_attention_dropout_path_runs():
+    torch.manual_seed(0)
+    B, T, H, D = 2, 32, 2, 8
+    qkv = torch.randn(B, T, 3 * H * D, requires_grad=True)
+    y = ops.fused_causal_attention(qkv, H, dropout_p=0.5, training=True)
+    assert y.shape == (B, T, H * D)
+    y.sum().backward()
+    assert qkv.grad is not None and torch.isfinite(qkv.grad).all()
+
+
+def test_model_math_and_fused_backends_agree():
+    from tiny_deepspeed_amd.models import GPTConfig, GPT2Model
+
+    torch.manual_seed(0)
+    cfg_f = GPTConfig(n_layer=2, n_head=2, n_embd=32, block_size=32,
+                      vocab_size=128, attention="fused")
+    m_f = GPT2Model(cfg_f)
+    torch.manual_seed(0)
+    cfg_m = GPTConfig(n_layer=2, n_head=2, n_embd=32, block_size=32,
+                      vocab_size=128, attention="math")
+    m_m = GPT2Model(cfg_m)
+    x = torch.randint(0, 128, (2, 16))
+    y = torch.randint(0, 128, (2, 16))
+    _, lf = m_f(x, y)
+    _, lm = m_m(x, y)
+    assert torch.allclose(lf, lm, atol=1e-5), (lf.item(), lm.item())

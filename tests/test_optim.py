@@ -1,6 +1,7 @@
 """Optimizer math vs torch.optim references (fp32), bf16 master-weight path,
 state_dict roundtrip, and the fixed global step counter."""
 
+import pytest
 import torch
 
 from tiny_deepspeed_amd import AdamW, SGD
@@ -107,3 +108,21 @@ def test_state_dict_roundtrip():
     for n in opt1.exp_avg:
         assert torch.allclose(opt1.exp_avg[n], opt2.exp_avg[n])
         assert torch.allclose(opt1.exp_avg_sq[n], opt2.exp_avg_sq[n])
+
+
+@pytest.mark.parametrize("nesterov", [False, True])
+def test_sgd_matches_torch_optim(nesterov):
+    torch.manual_seed(0)
+    p1 = torch.nn.Parameter(torch.randn(64, 32))
+    p2 = torch.nn.Parameter(p1.detach().clone())
+    ours = SGD([("p", p1)], lr=0.05, momentum=0.9, weight_decay=0.01,
+               nesterov=nesterov)
+    ref = torch.optim.SGD([p2], lr=0.05, momentum=0.9, weight_decay=0.01,
+                          nesterov=nesterov)
+    for _ in range(4):
+        g = torch.randn_like(p1)
+        p1.grad = g.clone()
+        p2.grad = g.clone()
+        ours.step()
+        ref.step()
+    assert torch.allclose(p1, p2, atol=1e-6), (p1 - p2).abs().max()
