@@ -10,11 +10,13 @@
 //             online softmax runs in exp2 space (log2e folded into the Q
 //             prescale); P is repacked into MFMA A-fragments with TWO
 //             v_permlane32_swap per k-slice (no LDS round trip); O = P V
-//             with V staged transposed. Saves per-row natural-log lse.
+//             with V consumed via ds_read_tr16_b64 hardware transpose
+//             reads from the row-major image. Saves per-row natural-log lse.
 //   backward: recompute-based two-kernel scheme (no atomics):
-//             dkv kernel owns a (NW*32)-key block and accumulates dK/dV over
-//             64-row Q tiles; dq kernel owns a (NW*32)-row Q block.
-//             delta = rowsum(dO*O) by a small wave-reduction kernel.
+//             the dq kernel (launched first) owns a (NW*32)-row Q block,
+//             computes delta = rowsum(dO*O) from its own fragments and
+//             publishes it; the dkv kernel owns a (NW*32)-key block and
+//             accumulates dK/dV over 64-row Q tiles, consuming delta.
 //             The 1/sqrt(D) on dS is folded into the dK/dQ epilogue.
 //
 // All global accesses are stride-parameterized so the kernels consume the
