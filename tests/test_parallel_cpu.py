@@ -81,3 +81,15 @@ def test_broadcast_bucketed_groups_small_tensors():
     results = run_distributed(W.broadcast_bucketed_roundtrip, world=2)
     for rank, ok in results.items():
         assert ok, f"rank {rank}"
+
+
+@pytest.mark.parametrize("strategy", ["zero1", "zero2", "zero3"])
+def test_loss_parity_world3(strategy):
+    """Odd world size: catches divisibility assumptions in partition /
+    owner maps / buckets before a real multi-GPU run."""
+    expected = W.single_device_losses()
+    results = run_distributed(W.train_strategy, world=3, args=(strategy,))
+    for rank, losses in results.items():
+        assert losses == pytest.approx(expected, abs=2e-4), (
+            f"{strategy} rank {rank}: {losses} != {expected}"
+        )
