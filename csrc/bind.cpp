@@ -429,8 +429,12 @@ std::vector<at::Tensor> attention_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                       c10::optional<at::Tensor> dk_out,
                                       c10::optional<at::Tensor> dv_out) {
   check_attn_tensor(q, "q");
+  check_attn_tensor(k, "k");
+  check_attn_tensor(v, "v");
   check_attn_tensor(o, "o");
   check_attn_tensor(dout, "dout");
+  TORCH_CHECK(q.strides() == k.strides() && q.strides() == v.strides(),
+              "q/k/v must share strides");
   TORCH_CHECK(o.strides() == dout.strides(), "o/dout must share strides");
   auto lsec = lse.contiguous();
   const long long B = q.size(0), H = q.size(1);

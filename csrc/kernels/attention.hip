@@ -43,7 +43,7 @@
 #define MFMA32(a, b, c) \
   __builtin_amdgcn_mfma_f32_32x32x16_bf16((a), (b), (c), 0, 0, 0)
 
-namespace {
+namespace tdsa {
 
 constexpr int KVB = 64;   // keys / rows per LDS tile
 constexpr int D = 64;     // head dim (all GPT-2 sizes)
@@ -713,7 +713,9 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
       dqp[crow(r, h32) * sd.t + dt * 32 + q32] = f2bf(dq_acc[dt][r] * scale);
 }
 
-}  // namespace
+}  // namespace tdsa
+
+using namespace tdsa;
 
 extern "C" {
 

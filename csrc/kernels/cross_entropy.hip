@@ -9,7 +9,7 @@
 
 #include <cstdlib>
 
-namespace {
+namespace tdsa {
 
 // merge two (m, s) logsumexp states (raw v_exp: exp(-inf) == 0, and the
 // branchless form keeps both scales well-defined for m == m2 == -inf)
@@ -168,7 +168,9 @@ __global__ void ce_bwd_kernel(const T* __restrict__ logits,
   }
 }
 
-}  // namespace
+}  // namespace tdsa
+
+using namespace tdsa;
 
 extern "C" {
 
@@ -176,6 +178,10 @@ hipError_t tdsa_ce_fwd(const void* logits, const long long* targets, float* lse,
                        float* loss_sum, int* n_valid, long long R, int V,
                        long long ignore_index, int is_bf16, hipStream_t stream) {
   const int block = 256;
+  // Row bases logits+row*V must stay 16B-aligned for the vector loads:
+  // reject V % W != 0 (mirrors the layernorm launcher guard; the python op
+  // routes such shapes to the composite path instead).
+  if (V % (is_bf16 ? 8 : 4)) return hipErrorInvalidValue;
   long long cap = 32768;
   if (const char* v = getenv("TDSA_CE_GRID")) cap = atoll(v);
   const int grid = (int)((R < cap) ? R : cap);
@@ -195,6 +201,7 @@ hipError_t tdsa_ce_bwd(const void* logits, const long long* targets,
                        float scale, long long ignore_index, int is_bf16,
                        hipStream_t stream) {
   const int block = 256;
+  if (V % (is_bf16 ? 8 : 4)) return hipErrorInvalidValue;  // see tdsa_ce_fwd
   long long cap = 32768;
   if (const char* v = getenv("TDSA_CE_GRID")) cap = atoll(v);
   const int grid = (int)((R < cap) ? R : cap);

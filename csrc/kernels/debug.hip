@@ -8,7 +8,7 @@
 //               the staging/swizzle/fragment-read path without MFMA.
 #include "common.h"
 
-namespace {
+namespace tdsa {
 
 typedef __attribute__((ext_vector_type(8))) short bfrag;
 
@@ -126,7 +126,9 @@ __global__ void dbg_tr16_kernel(const bf16* __restrict__ in,
       }
 }
 
-}  // namespace
+}  // namespace tdsa
+
+using namespace tdsa;
 
 extern "C" {
 

@@ -6,7 +6,7 @@
 // Guideline 13; grid-stride with a capped grid, Guideline 11).
 #include "common.h"
 
-namespace {
+namespace tdsa {
 
 constexpr float GC0 = 0.7978845608028654f;  // sqrt(2/pi)
 constexpr float GC1 = 0.044715f;
@@ -119,7 +119,9 @@ __global__ void cast_from_f32(const float* __restrict__ in, T* __restrict__ out,
   for (long long i = global_tid(); i < n; i += stride) out[i] = (T)in[i];
 }
 
-}  // namespace
+}  // namespace tdsa
+
+using namespace tdsa;
 
 extern "C" {
 

@@ -7,7 +7,7 @@
 // (bf16 atomics would lose small contributions).
 #include "common.h"
 
-namespace {
+namespace tdsa {
 
 // vector type of width W over T
 template <typename T, int W> struct VecOf;
@@ -59,7 +59,9 @@ __global__ void emb_bwd_kernel(const T* __restrict__ dy,
   }
 }
 
-}  // namespace
+}  // namespace tdsa
+
+using namespace tdsa;
 
 extern "C" {
 

@@ -11,7 +11,7 @@
 
 #include <cstdlib>
 
-namespace {
+namespace tdsa {
 
 template <typename T> struct VecTraits;
 template <> struct VecTraits<float> {
@@ -220,7 +220,9 @@ __global__ void ln_bwd_dwdb_kernel(const float* __restrict__ pdw,
   }
 }
 
-}  // namespace
+}  // namespace tdsa
+
+using namespace tdsa;
 
 extern "C" {
 

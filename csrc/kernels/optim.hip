@@ -8,7 +8,7 @@
 // 1-based optimizer step.
 #include "common.h"
 
-namespace {
+namespace tdsa {
 
 template <typename PT, typename GT>
 __global__ void adamw_kernel(PT* __restrict__ param, const GT* __restrict__ grad,
@@ -200,7 +200,9 @@ __global__ void sgd_multi_kernel(const SgdTensorDesc* __restrict__ descs,
   }
 }
 
-}  // namespace
+}  // namespace tdsa
+
+using namespace tdsa;
 
 extern "C" {
 

@@ -30,7 +30,22 @@ def main():
     print(f"{'total_ms':>9} {'calls':>6} {'avg_us':>9} {'%':>5} "
           f"{'vgpr':>5} {'lds':>6}  kernel")
     for name, n, ms, us, vgpr, lds in rows:
-        short = name.split("(")[0]
+        # strip return type / namespace noise so every row is attributable
+        # (a bare split("(") turned "(anonymous namespace)::k<...>(args)"
+        # into "void " — the round-1 unnamed rows)
+        short = name.replace("(anonymous namespace)::", "")
+        short = short.replace("tdsa::", "")
+        if short.startswith("void "):
+            short = short[5:]
+        depth = 0
+        for i, c in enumerate(short):  # split at the ARG paren, not template ones
+            if c == "<":
+                depth += 1
+            elif c == ">":
+                depth -= 1
+            elif c == "(" and depth == 0:
+                short = short[:i]
+                break
         if len(short) > 80:
             short = short[:80]
         print(f"{ms:9.2f} {n:6d} {us:9.1f} {100*ms/total:5.1f} "

@@ -19,6 +19,9 @@ class _LayerNormResFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, res, weight, bias, module):
         ctx.module = module
+        # don't materialize an all-zeros dh when h is unused (ln_f discards
+        # it) — backward treats dh=None as the plain no-fold path
+        ctx.set_materialize_grads(False)
         h, y, mean, rstd = module.forward_res_callback(x, res, weight, bias)
         ctx.save_for_backward(h, mean, rstd)
         return h, y
@@ -26,6 +29,8 @@ class _LayerNormResFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dh, dy):
         h, mean, rstd = ctx.saved_tensors
+        if dy is None:  # y unused: gradient flows through h alone
+            return dh, dh, None, None, None
         dx, dw, db = ctx.module.backward_callback(dy, h, mean, rstd, dh=dh)
         return dx, dx, dw, db, None
 

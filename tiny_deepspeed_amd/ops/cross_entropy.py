@@ -12,9 +12,17 @@ import torch
 from . import _ext
 
 
+def _kernel_supported(logits2d):
+    """The CDNA4 kernels vector-load rows: V must keep row bases 16B-aligned
+    (V % 8 bf16 / V % 4 fp32 — guarded in tdsa_ce_fwd/bwd). Off-width vocabs
+    (e.g. unpadded 50257) use the composite path."""
+    w = 8 if logits2d.dtype == torch.bfloat16 else 4
+    return logits2d.shape[-1] % w == 0
+
+
 def cross_entropy_fwd(logits2d, targets, ignore_index=-100, tuner=None):
     """Returns (loss_sum[fp32 scalar], lse[rows fp32], n_valid[int64 scalar])."""
-    if _ext.use_native(logits2d):
+    if _ext.use_native(logits2d) and _kernel_supported(logits2d):
         return _ext.get_ext().cross_entropy_fwd(
             logits2d.contiguous(), targets.contiguous(), ignore_index
         )
@@ -29,7 +37,7 @@ def cross_entropy_fwd(logits2d, targets, ignore_index=-100, tuner=None):
 
 
 def cross_entropy_bwd(dloss, logits2d, targets, lse, n_valid, ignore_index=-100, tuner=None):
-    if _ext.use_native(logits2d):
+    if _ext.use_native(logits2d) and _kernel_supported(logits2d):
         return _ext.get_ext().cross_entropy_bwd(
             logits2d.contiguous(), targets.contiguous(), lse,
             float(dloss), int(n_valid), ignore_index,

@@ -61,6 +61,9 @@ class CommContext:
         }
         self._works = {REDUCE: [], GATHER: []}
         self._keepalive = []
+        # gloo flat broadcasts whose scatter-back is deferred to sync():
+        # (work, flat, bucket, numels) per channel
+        self._pending_flat = {REDUCE: [], GATHER: []}
 
     # ------------------------------------------------------------------ #
     def _launch(self, tensors, collective, channel):
@@ -86,6 +89,8 @@ class CommContext:
             for w in works:
                 w.wait()
             works.clear()
+        for channel in (REDUCE, GATHER):
+            self._drain_flat(channel)
         self._keepalive.clear()
 
     def wait_gather(self):
@@ -96,6 +101,18 @@ class CommContext:
         for w in self._works[GATHER]:
             w.wait()
         self._works[GATHER].clear()
+        self._drain_flat(GATHER)
+
+    def _drain_flat(self, channel):
+        """Complete deferred gloo flat broadcasts: wait + scatter-back."""
+        for work, flat, bucket, numels, src in self._pending_flat[channel]:
+            work.wait()
+            if self.rank != src:
+                off = 0
+                for t, n in zip(bucket, numels):
+                    t.view(-1).copy_(flat[off:off + n])
+                    off += n
+        self._pending_flat[channel].clear()
 
     def keep_until_sync(self, t):
         """Pin a tensor's host reference until the next sync() (gloo path;
@@ -180,33 +197,29 @@ class CommContext:
     def _broadcast_flat(self, bucket, src, channel):
         stream = self.streams[channel]
         numels = [t.numel() for t in bucket]
-
-        def body(async_op=False):
-            if self.rank == src:
-                flat = torch.cat([t.reshape(-1) for t in bucket])
-            else:
-                flat = torch.empty(sum(numels), dtype=bucket[0].dtype,
-                                   device=bucket[0].device)
-            work = dist.broadcast(flat, src=src, group=self.pg[channel],
-                                  async_op=async_op)
-            if async_op and work is not None:
-                work.wait()  # gloo path: complete before scatter-back
-            if self.rank != src:
-                off = 0
-                for t, n in zip(bucket, numels):
-                    t.view(-1).copy_(flat[off:off + n])
-                    off += n
-            return flat
-
+        if self.rank == src:
+            flat = torch.cat([t.reshape(-1) for t in bucket])
+        else:
+            flat = torch.empty(sum(numels), dtype=bucket[0].dtype,
+                               device=bucket[0].device)
         if stream is not None:
             stream.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(stream):
-                flat = body()
+                dist.broadcast(flat, src=src, group=self.pg[channel])
+                if self.rank != src:
+                    off = 0
+                    for t, n in zip(bucket, numels):
+                        t.view(-1).copy_(flat[off:off + n])
+                        off += n
                 flat.record_stream(stream)
                 for t in bucket:
                     t.record_stream(stream)
         else:
-            self._keepalive.append(body(async_op=True))
+            # gloo: stay async until sync()/wait_gather() — the scatter-back
+            # to non-src ranks is deferred with the completion handle
+            work = dist.broadcast(flat, src=src, group=self.pg[channel],
+                                  async_op=True)
+            self._pending_flat[channel].append((work, flat, bucket, numels, src))
 
     def all_reduce_scalar_avg(self, t):
         """Synchronous scalar average (loss logging)."""
