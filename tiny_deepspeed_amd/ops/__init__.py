@@ -19,7 +19,7 @@ from .gelu import gelu, gelu_fwd, gelu_bwd
 from .attention import causal_attention, fused_causal_attention
 from .cross_entropy import cross_entropy
 from .optim_ops import adamw_step, sgd_step
-from .autotuner import RuntimeAutoTuner
+from .autotuner import RuntimeAutoTuner, default_tuner
 from ._ext import ext_available, get_ext
 from .utils import acc_dtype
 
@@ -31,7 +31,7 @@ __all__ = [
     "causal_attention", "fused_causal_attention",
     "cross_entropy",
     "adamw_step", "sgd_step",
-    "RuntimeAutoTuner",
+    "RuntimeAutoTuner", "default_tuner",
     "ext_available", "get_ext",
     "acc_dtype",
 ]

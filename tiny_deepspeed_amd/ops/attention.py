@@ -16,6 +16,7 @@ import math
 import torch
 
 from . import _ext
+from .autotuner import default_tuner
 
 
 def _composite_fwd(q, k, v, scale):
@@ -58,13 +59,67 @@ def _kernel_supported(q):
     )
 
 
+# --- autotuner candidate pairs (identical signatures & output contracts;
+# the packed variants write into caller-provided layout views so both
+# implementations pay their true cost including any transpose copies) ------
+def attn_fwd_hip(q, k, v, scale):
+    return _ext.get_ext().attention_fwd(
+        q.contiguous(), k.contiguous(), v.contiguous(), scale
+    )
+
+
+def attn_fwd_composite(q, k, v, scale):
+    return _composite_fwd(q, k, v, scale)
+
+
+def attn_bwd_hip(q, k, v, o, lse, do, scale):
+    return _ext.get_ext().attention_bwd(
+        q.contiguous(), k.contiguous(), v.contiguous(),
+        o.contiguous(), lse, do.contiguous(), scale,
+    )
+
+
+def attn_bwd_composite(q, k, v, o, lse, do, scale):
+    return _composite_bwd(q, k, v, o, lse, do, scale)
+
+
+def attn_fwd_packed_hip(q, k, v, scale, o_view):
+    return _ext.get_ext().attention_fwd(q, k, v, scale, o_view)
+
+
+def attn_fwd_packed_composite(q, k, v, scale, o_view):
+    o, lse = _composite_fwd(q.contiguous(), k.contiguous(), v.contiguous(),
+                            scale)
+    o_view.copy_(o)
+    return o_view, lse
+
+
+def attn_bwd_packed_hip(q, k, v, o, lse, do, scale, dq, dk, dv):
+    return _ext.get_ext().attention_bwd(q, k, v, o, lse, do, scale,
+                                        dq, dk, dv)
+
+
+def attn_bwd_packed_composite(q, k, v, o, lse, do, scale, dq, dk, dv):
+    dq_c, dk_c, dv_c = _composite_bwd(
+        q.contiguous(), k.contiguous(), v.contiguous(),
+        o.contiguous(), lse, do.contiguous(), scale)
+    dq.copy_(dq_c)
+    dk.copy_(dk_c)
+    dv.copy_(dv_c)
+    return dq, dk, dv
+
+
 class _CausalAttentionFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale):
         if _ext.use_native(q) and _kernel_supported(q):
-            o, lse = _ext.get_ext().attention_fwd(
-                q.contiguous(), k.contiguous(), v.contiguous(), scale
-            )
+            tuner = default_tuner()
+            if tuner is not None:
+                o, lse = tuner.choose("attn_fwd",
+                                      [attn_fwd_hip, attn_fwd_composite],
+                                      q, k, v, scale)
+            else:
+                o, lse = attn_fwd_hip(q, k, v, scale)
         else:
             o, lse = _composite_fwd(q, k, v, scale)
         ctx.save_for_backward(q, k, v, o, lse)
@@ -75,10 +130,13 @@ class _CausalAttentionFn(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         if _ext.use_native(q) and _kernel_supported(q):
-            dq, dk, dv = _ext.get_ext().attention_bwd(
-                q.contiguous(), k.contiguous(), v.contiguous(),
-                o.contiguous(), lse, do.contiguous(), ctx.scale,
-            )
+            tuner = default_tuner()
+            if tuner is not None:
+                dq, dk, dv = tuner.choose(
+                    "attn_bwd", [attn_bwd_hip, attn_bwd_composite],
+                    q, k, v, o, lse, do, ctx.scale)
+            else:
+                dq, dk, dv = attn_bwd_hip(q, k, v, o, lse, do, ctx.scale)
         else:
             dq, dk, dv = _composite_bwd(q, k, v, o, lse, do, ctx.scale)
         return dq, dk, dv, None
@@ -111,7 +169,14 @@ class _FusedQKVAttentionFn(torch.autograd.Function):
         if _ext.use_native(qkv) and _kernel_supported(q):
             y = torch.empty(B, T, E, dtype=qkv.dtype, device=qkv.device)
             o_view = y.view(B, T, n_head, D).permute(0, 2, 1, 3)
-            o, lse = _ext.get_ext().attention_fwd(q, k, v, scale, o_view)
+            tuner = default_tuner()
+            if tuner is not None:
+                o, lse = tuner.choose(
+                    "attn_fwd_packed",
+                    [attn_fwd_packed_hip, attn_fwd_packed_composite],
+                    q, k, v, scale, o_view)
+            else:
+                o, lse = attn_fwd_packed_hip(q, k, v, scale, o_view)
             ctx.native = True
         else:
             o, lse = _composite_fwd(q.contiguous(), k.contiguous(),
@@ -136,8 +201,15 @@ class _FusedQKVAttentionFn(torch.autograd.Function):
             dq = dqkv4[:, :, 0].permute(0, 2, 1, 3)
             dk = dqkv4[:, :, 1].permute(0, 2, 1, 3)
             dv = dqkv4[:, :, 2].permute(0, 2, 1, 3)
-            _ext.get_ext().attention_bwd(q, k, v, o_view, lse, do_view,
-                                         ctx.scale, dq, dk, dv)
+            tuner = default_tuner()
+            if tuner is not None:
+                tuner.choose("attn_bwd_packed",
+                             [attn_bwd_packed_hip, attn_bwd_packed_composite],
+                             q, k, v, o_view, lse, do_view, ctx.scale,
+                             dq, dk, dv)
+            else:
+                attn_bwd_packed_hip(q, k, v, o_view, lse, do_view, ctx.scale,
+                                    dq, dk, dv)
         else:
             dq, dk, dv = _composite_bwd(
                 q.contiguous(), k.contiguous(), v.contiguous(),

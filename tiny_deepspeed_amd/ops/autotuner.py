@@ -10,6 +10,7 @@ Candidates are callables taking identical args. The default (index 0) is the
 hand-written HIP kernel; later entries are library/torch fallbacks.
 """
 
+import os
 import time
 
 import torch
@@ -47,7 +48,8 @@ class RuntimeAutoTuner:
         return out
 
     def _time_one(self, fn, args, kwargs):
-        if args and isinstance(args[0], torch.Tensor) and args[0].is_cuda:
+        on_gpu = any(isinstance(a, torch.Tensor) and a.is_cuda for a in args)
+        if on_gpu:
             start = torch.cuda.Event(enable_timing=True)
             stop = torch.cuda.Event(enable_timing=True)
             for _ in range(self.warmup):
@@ -81,3 +83,29 @@ class RuntimeAutoTuner:
     def final_tune(self):
         """Freeze the choices made so far (reference parity: ``final_tune``)."""
         self.finalized = True
+
+    def choices(self):
+        """{(op_name, shape_key): chosen callable __name__} — introspection
+        for tests and for logging which implementation won per shape."""
+        return {k: fn.__name__ for k, fn in self._best.items()}
+
+
+_DEFAULT_TUNER = None
+
+
+def default_tuner():
+    """Process-wide tuner used by the op layer when the module didn't pass
+    one. Live on GPU by default — every op with >1 candidate implementation
+    dispatches through measured choice (the reference routes every op
+    through a candidate list, /root/reference/tiny_deepspeed/core/module/
+    ops/linear.py:9-17). ``TDSA_AUTOTUNE=0`` disables (candidate[0], the
+    hand-written kernel, runs unconditionally); on CPU there is nothing to
+    tune (single torch candidate) so None is returned."""
+    global _DEFAULT_TUNER
+    if os.environ.get("TDSA_AUTOTUNE", "1") == "0":
+        return None
+    if not torch.cuda.is_available():
+        return None
+    if _DEFAULT_TUNER is None:
+        _DEFAULT_TUNER = RuntimeAutoTuner()
+    return _DEFAULT_TUNER

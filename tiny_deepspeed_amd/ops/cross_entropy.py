@@ -5,11 +5,14 @@ Here: a fused CDNA4 kernel pair — forward computes per-row max/logsumexp and
 the NLL in one HBM pass over the (B*T, 50304) logits; backward writes
 dlogits = (softmax - onehot) * scale in one pass. fp32 accumulation,
 logits dtype in/out. Mean reduction over rows (ignore_index supported).
+Dispatch is {CDNA4 kernel, torch composite} through the runtime autotuner
+(reference candidate-list architecture, ``ops/linear.py:9-17``).
 """
 
 import torch
 
 from . import _ext
+from .autotuner import default_tuner
 
 
 def _kernel_supported(logits2d):
@@ -20,12 +23,13 @@ def _kernel_supported(logits2d):
     return logits2d.shape[-1] % w == 0
 
 
-def cross_entropy_fwd(logits2d, targets, ignore_index=-100, tuner=None):
-    """Returns (loss_sum[fp32 scalar], lse[rows fp32], n_valid[int64 scalar])."""
-    if _ext.use_native(logits2d) and _kernel_supported(logits2d):
-        return _ext.get_ext().cross_entropy_fwd(
-            logits2d.contiguous(), targets.contiguous(), ignore_index
-        )
+def ce_fwd_hip(logits2d, targets, ignore_index):
+    return _ext.get_ext().cross_entropy_fwd(
+        logits2d.contiguous(), targets.contiguous(), ignore_index
+    )
+
+
+def ce_fwd_torch(logits2d, targets, ignore_index):
     lf = logits2d.float()
     m = lf.max(dim=-1).values
     lse = m + (lf - m.unsqueeze(-1)).exp().sum(dim=-1).log()
@@ -36,12 +40,25 @@ def cross_entropy_fwd(logits2d, targets, ignore_index=-100, tuner=None):
     return losses.sum(), lse, valid.sum()
 
 
-def cross_entropy_bwd(dloss, logits2d, targets, lse, n_valid, ignore_index=-100, tuner=None):
-    if _ext.use_native(logits2d) and _kernel_supported(logits2d):
-        return _ext.get_ext().cross_entropy_bwd(
-            logits2d.contiguous(), targets.contiguous(), lse,
-            float(dloss), int(n_valid), ignore_index,
-        )
+def cross_entropy_fwd(logits2d, targets, ignore_index=-100, tuner=None):
+    """Returns (loss_sum[fp32 scalar], lse[rows fp32], n_valid[int64 scalar])."""
+    if not (_ext.use_native(logits2d) and _kernel_supported(logits2d)):
+        return ce_fwd_torch(logits2d, targets, ignore_index)
+    tuner = tuner if tuner is not None else default_tuner()
+    if tuner is not None:
+        return tuner.choose("ce_fwd", [ce_fwd_hip, ce_fwd_torch],
+                            logits2d, targets, ignore_index)
+    return ce_fwd_hip(logits2d, targets, ignore_index)
+
+
+def ce_bwd_hip(dloss, logits2d, targets, lse, n_valid, ignore_index):
+    return _ext.get_ext().cross_entropy_bwd(
+        logits2d.contiguous(), targets.contiguous(), lse,
+        float(dloss), int(n_valid), ignore_index,
+    )
+
+
+def ce_bwd_torch(dloss, logits2d, targets, lse, n_valid, ignore_index):
     lf = logits2d.float()
     soft = (lf - lse.unsqueeze(-1)).exp()
     valid = (targets != ignore_index).unsqueeze(1)
@@ -52,6 +69,16 @@ def cross_entropy_bwd(dloss, logits2d, targets, lse, n_valid, ignore_index=-100,
     scale = float(dloss) / max(int(n_valid), 1)
     dlogits = torch.where(valid, soft * scale, torch.zeros_like(soft))
     return dlogits.to(logits2d.dtype)
+
+
+def cross_entropy_bwd(dloss, logits2d, targets, lse, n_valid, ignore_index=-100, tuner=None):
+    if not (_ext.use_native(logits2d) and _kernel_supported(logits2d)):
+        return ce_bwd_torch(dloss, logits2d, targets, lse, n_valid, ignore_index)
+    tuner = tuner if tuner is not None else default_tuner()
+    if tuner is not None:
+        return tuner.choose("ce_bwd", [ce_bwd_hip, ce_bwd_torch],
+                            dloss, logits2d, targets, lse, n_valid, ignore_index)
+    return ce_bwd_hip(dloss, logits2d, targets, lse, n_valid, ignore_index)
 
 
 class _CrossEntropyFn(torch.autograd.Function):

@@ -1,25 +1,28 @@
 """Linear (affine) op functions: forward, input-grad, weight-grad, bias-grad.
 
 Capability parity with the reference ops
-(``/root/reference/tiny_deepspeed/core/module/ops/linear.py:9-75``).
+(``/root/reference/tiny_deepspeed/core/module/ops/linear.py:9-75``), and the
+same candidate-list dispatch architecture (reference ``ops/linear.py:9-17``):
+each op owns a list of implementations and routes through the runtime
+autotuner, which times them per shape and caches the winner.
 
-MI355X design note: plain GEMMs belong to the library path — torch.matmul on
-ROCm dispatches to hipBLASLt/rocBLAS which drive the MFMA matrix cores
-directly, so these four functions are thin shape adapters around matmul with
-the right transpose forms:
+  fwd : out[M,N] = x[M,K] @ W[N,K]^T (+ b)      (NT GEMM — hipBLASLt)
+  dX  : dx[M,K]  = dy[M,N] @ W[N,K]             (NN GEMM — hipBLASLt)
+  dW  : dw[N,K]  = dy[M,N]^T @ x[M,K]           (TN GEMM — {CDNA4 MFMA
+                                                 kernel, hipBLASLt}, tuned)
+  db  : db[N]    = sum_M dy[M,N]                ({CDNA4 column_sum, torch})
 
-  fwd : out[M,N] = x[M,K] @ W[N,K]^T (+ b)      (NT GEMM)
-  dX  : dx[M,K]  = dy[M,N] @ W[N,K]             (NN GEMM)
-  dW  : dw[N,K]  = dy[M,N]^T @ x[M,K]           (TN GEMM)
-  db  : db[N]    = sum_M dy[M,N]                (column reduce, HIP kernel)
-
-The *fused* hot ops (layernorm, gelu, attention, cross-entropy, optimizers,
-embedding) are hand-written CDNA4 kernels — see the sibling modules.
+MI355X design note: the plain NT/NN GEMMs stay on the library path —
+measured at the practical MFMA ceiling (profiles/gemm_shapes_hipblaslt.txt,
+fwd NT 1436 TF/s) — while dW (TN, the weakest library shape) carries a
+hand-written split-M MFMA kernel as a tuner candidate so the question is
+settled by measurement per shape, not by assertion.
 """
 
 import torch
 
 from . import _ext
+from .autotuner import default_tuner
 
 
 def _flatten_batch(t):
@@ -38,14 +41,56 @@ def linear_input_grad(dy, weight, tuner=None):
     return torch.matmul(dy, weight)
 
 
+# --- dW candidates ---------------------------------------------------------
+def dw_library(dy2, x2):
+    """hipBLASLt TN GEMM via torch.matmul."""
+    return torch.matmul(dy2.t(), x2)
+
+
+def dw_hip(dy2, x2):
+    """Hand-written CDNA4 split-M MFMA TN kernel (csrc/kernels/gemm_tn.hip)."""
+    return _ext.get_ext().gemm_tn(dy2.contiguous(), x2.contiguous())
+
+
+def _dw_hip_supported(dy2, x2):
+    if not _ext.ext_available() or not hasattr(_ext.get_ext(), "gemm_tn"):
+        return False
+    # kernel contract: bf16, M % 64 == 0, N/K % 16 == 0
+    return (
+        dy2.dtype == torch.bfloat16
+        and x2.dtype == torch.bfloat16
+        and dy2.shape[0] % 64 == 0
+        and dy2.shape[1] % 16 == 0
+        and x2.shape[1] % 16 == 0
+    )
+
+
 def linear_weight_grad(dy, x, tuner=None):
     dy2 = _flatten_batch(dy)
     x2 = _flatten_batch(x)
-    return torch.matmul(dy2.t(), x2)
+    candidates = [dw_library]
+    if dy2.is_cuda and _ext.use_native(dy2) and _dw_hip_supported(dy2, x2):
+        candidates.insert(0, dw_hip)
+    tuner = tuner if tuner is not None else default_tuner()
+    if tuner is not None and dy2.is_cuda:
+        return tuner.choose("linear_dw", candidates, dy2, x2)
+    return candidates[0](dy2, x2)
+
+
+# --- db candidates ---------------------------------------------------------
+def db_hip(dy2):
+    return _ext.get_ext().column_sum(dy2.contiguous())
+
+
+def db_torch(dy2):
+    return dy2.sum(dim=0)
 
 
 def linear_bias_grad(dy, tuner=None):
     dy2 = _flatten_batch(dy)
-    if _ext.use_native(dy2):
-        return _ext.get_ext().column_sum(dy2.contiguous())
-    return dy2.sum(dim=0)
+    if not _ext.use_native(dy2):
+        return db_torch(dy2)
+    tuner = tuner if tuner is not None else default_tuner()
+    if tuner is not None:
+        return tuner.choose("linear_db", [db_hip, db_torch], dy2)
+    return db_hip(dy2)
