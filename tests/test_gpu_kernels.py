@@ -295,7 +295,11 @@ def test_adamw_multi_matches_single_gpu():
         _close(o1.master[n], o2.master[n], 1e-6, "master multi vs single")
 
 
-@pytest.mark.parametrize("N", [1024, 1280, 1600])  # medium/large/xl widths
+@pytest.mark.parametrize("N", [1024, 1280, 1600,
+                               # beyond the register-cache bound: the wide
+                               # two-pass kernels (4096 bf16 is the last
+                               # narrow width; 8192/16384 chain to wide)
+                               8192, 16384])
 def test_layernorm_model_widths_gpu(N):
     torch.manual_seed(0)
     M = 256
@@ -312,6 +316,37 @@ def test_layernorm_model_widths_gpu(N):
     ref2 = torch.nn.functional.layer_norm(ref_h, (N,), w.float(), b.float())
     _close(h, ref_h, 2e-2, f"ln res h N={N}")
     _close(y2, ref2, 2e-2, f"ln res y N={N}")
+
+
+@pytest.mark.parametrize("N,dtype", [(8192, torch.bfloat16),
+                                     (8192, torch.float32),
+                                     (4096, torch.float32)])
+def test_layernorm_wide_bwd_gpu(N, dtype):
+    """Wide fallback backward (N beyond the register-cache bound — the
+    round-1 launcher returned hipErrorInvalidValue here)."""
+    torch.manual_seed(1)
+    M = 512
+    tol = 2e-2 if dtype == torch.bfloat16 else 1e-4
+    x = torch.randn(M, N, device="cuda", dtype=dtype)
+    w = torch.randn(N, device="cuda", dtype=dtype)
+    b = torch.randn(N, device="cuda", dtype=dtype)
+    y, mean, rstd = ops.layernorm_fwd(x, w, b)
+    dy = torch.randn(M, N, device="cuda", dtype=dtype)
+    dh = torch.randn(M, N, device="cuda", dtype=dtype)
+    xf = x.float().requires_grad_(True)
+    wf = w.float().requires_grad_(True)
+    bf = b.float().requires_grad_(True)
+    ref = torch.nn.functional.layer_norm(xf, (N,), wf, bf)
+    _close(y, ref, tol, f"ln wide fwd N={N}")
+    ref.backward(dy.float())
+    dx, ws = ops.layernorm_dx(dy, x, w, mean, rstd)
+    dw, db = ops.layernorm_dwdb(ws, dtype=dtype)
+    _close(dx, xf.grad, tol * 4, f"ln wide dx N={N}")
+    _close(dw, wf.grad, tol * 4, f"ln wide dw N={N}")
+    _close(db, bf.grad, tol * 4, f"ln wide db N={N}")
+    # fused dh fold-in
+    dx2, _ = ops.layernorm_dx(dy, x, w, mean, rstd, dh=dh)
+    _close(dx2, xf.grad + dh.float(), tol * 4, f"ln wide dx+dh N={N}")
 
 
 def test_layernorm_fused_bwd_gpu():

@@ -40,7 +40,7 @@ class _ZeroOptimMixin:
         self.comm.sync()
 
     def post_step(self):
-        if self.broadcast_params_after_step and self.comm.world_size > 1:
+        if self.broadcast_params_after_step and not self.comm._inactive():
             # refresh replicas: bucketed async broadcasts from each owner,
             # identical order on all ranks, then one stream wait
             self.comm.broadcast_bucketed(

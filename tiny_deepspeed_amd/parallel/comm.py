@@ -32,6 +32,8 @@ On CPU (gloo backend, used by the no-GPU tests) there are no streams;
 collectives run async_op=True and sync()/wait_gather() wait the handles.
 """
 
+import os
+
 import torch
 import torch.distributed as dist
 
@@ -42,17 +44,24 @@ GATHER = "gather"
 class CommContext:
     def __init__(self, process_group=None, use_comm_stream=True):
         self.initialized = dist.is_available() and dist.is_initialized()
+        # TDSA_COMM_FORCE=1: run every collective even at world size 1.
+        # Lets 1-GPU leases exercise the real RCCL enqueue / comm-stream
+        # ordering / record_stream lifetime path that the world-1
+        # early-returns below would otherwise skip (hardware evidence for
+        # the multi-GPU machinery before an 8-GPU node exists).
+        self.force = os.environ.get("TDSA_COMM_FORCE", "0") == "1"
         self.pg = {REDUCE: process_group, GATHER: process_group}
         if self.initialized:
             self.rank = dist.get_rank(process_group)
             self.world_size = dist.get_world_size(process_group)
-            if self.world_size > 1 and process_group is None:
+            if (self.world_size > 1 or self.force) and process_group is None:
                 # second communicator for the gather channel (must be
                 # constructed collectively, identical on all ranks)
                 self.pg[GATHER] = dist.new_group(backend=dist.get_backend())
         else:
             self.rank = 0
             self.world_size = 1
+            self.force = False
         self.is_cuda = torch.cuda.is_available()
         use_streams = self.is_cuda and use_comm_stream
         self.streams = {
@@ -66,6 +75,12 @@ class CommContext:
         self._pending_flat = {REDUCE: [], GATHER: []}
 
     # ------------------------------------------------------------------ #
+    def _inactive(self):
+        """True when collectives should no-op (world 1, unless forced)."""
+        if not self.initialized:
+            return True
+        return self.world_size == 1 and not self.force
+
     def _launch(self, tensors, collective, channel):
         stream = self.streams[channel]
         if stream is not None:
@@ -123,7 +138,7 @@ class CommContext:
     # --- collectives ---------------------------------------------------- #
     def all_reduce_avg(self, t):
         """Async in-place average-all-reduce; returns t."""
-        if not self.initialized or self.world_size == 1:
+        if self._inactive():
             return t
 
         def run(async_op=False):
@@ -136,7 +151,7 @@ class CommContext:
 
     def reduce_avg_to(self, t, owner):
         """Async in-place average-reduce to `owner`; valid only there."""
-        if not self.initialized or self.world_size == 1:
+        if self._inactive():
             return t
 
         def run(async_op=False):
@@ -149,7 +164,7 @@ class CommContext:
 
     def broadcast(self, t, src, channel=REDUCE):
         """Async in-place broadcast from `src`; returns t."""
-        if not self.initialized or self.world_size == 1:
+        if self._inactive():
             return t
 
         def run(async_op=False):
@@ -170,7 +185,7 @@ class CommContext:
         census has ~100 sub-1MB payloads whose per-collective launch latency
         dominates their transfer time. Large tensors broadcast directly
         (a flatten round trip would double their local traffic)."""
-        if not self.initialized or self.world_size == 1:
+        if self._inactive():
             return
         groups = []  # (src, [tensors]) of consecutive small same-src tensors
         for t, src in tensors_with_src:
@@ -223,7 +238,7 @@ class CommContext:
 
     def all_reduce_scalar_avg(self, t):
         """Synchronous scalar average (loss logging)."""
-        if not self.initialized or self.world_size == 1:
+        if self._inactive():
             return t
         dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.pg[REDUCE])
         t.div_(self.world_size)

@@ -51,7 +51,7 @@ class _GatherMixin:
         if param is None:
             return None
         comm = self._comm
-        if comm.world_size == 1:
+        if comm._inactive():
             return param
         if comm.rank == param._tdsa_owner:
             comm.gather_broadcast(param.data, src=param._tdsa_owner)
