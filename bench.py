@@ -139,6 +139,9 @@ def main():
         dist.barrier()
     if have_gpu:
         torch.cuda.synchronize()
+        # report steady-state peak memory: warmup includes one-time autotuner
+        # candidate timing whose composite transients never recur
+        torch.cuda.reset_peak_memory_stats()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         step()

@@ -182,6 +182,68 @@ def grad_accumulation(rank, world):
     return lin.weight.grad.clone()
 
 
+def _build_zero(rank, world, strategy):
+    import tiny_deepspeed_amd as tdsa
+    from tiny_deepspeed_amd.models import GPT2Model
+
+    torch.manual_seed(0)
+    model = GPT2Model(make_cfg())
+    with torch.device("meta"):
+        meta = GPT2Model(make_cfg())
+    parts, _ = tdsa.partition_tensors(
+        OrderedDict(meta.named_parameters()), ["cpu"] * world
+    )
+    wrap = {"zero1": tdsa.Zero1, "zero2": tdsa.Zero2}[strategy]
+    optc = {"zero1": tdsa.Zero1AdamW, "zero2": tdsa.Zero2AdamW}[strategy]
+    model = wrap(model, parts)
+    opt = optc(model.named_parameters(), lr=1e-3, weight_decay=0.01,
+               param_part_table=parts, ranks_map=["cpu"] * world)
+    return model, opt
+
+
+def _run_steps(model, opt, seeds):
+    losses = []
+    for s in seeds:
+        g = torch.Generator().manual_seed(s)
+        x = torch.randint(0, CFG["vocab_size"], (2, CFG["block_size"]),
+                          generator=g)
+        y = torch.randint(0, CFG["vocab_size"], (2, CFG["block_size"]),
+                          generator=g)
+        model.require_backward_grad_sync = True
+        _, loss = model(x, y)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    return losses
+
+
+def zero_ckpt_train_save(rank, world, strategy, path):
+    """Train 2 steps, save the per-rank shard, continue 2 more steps;
+    return the continuation losses (the resharded-resume target)."""
+    from tiny_deepspeed_amd.utils.checkpoint import save_checkpoint
+
+    model, opt = _build_zero(rank, world, strategy)
+    _run_steps(model, opt, [0, 1])
+    save_checkpoint(path, model, opt, step=2, rank=rank, world_size=world)
+    import torch.distributed as dist
+    dist.barrier()
+    return _run_steps(model, opt, [10, 11])
+
+
+def zero_ckpt_resume(rank, world, strategy, path):
+    """Fresh build under a DIFFERENT world size, merge-load every shard,
+    run the same continuation steps."""
+    from tiny_deepspeed_amd.utils.checkpoint import load_checkpoint
+
+    model, opt = _build_zero(rank, world, strategy)
+    step, report = load_checkpoint(path, model, opt, rank=rank,
+                                   return_report=True)
+    assert step == 2
+    assert not report.mismatched and not report.unexpected, repr(report)
+    assert opt.t == 2
+    return _run_steps(model, opt, [10, 11])
+
+
 def broadcast_bucketed_roundtrip(rank, world):
     """Mixed owners/sizes through comm.broadcast_bucketed: every rank ends
     with the owner's values (small tensors ride flat buckets)."""

@@ -49,6 +49,63 @@ def test_checkpoint_roundtrip(tmp_path):
     assert new_losses == ref_losses
 
 
+def test_mismatched_shapes_reported(tmp_path):
+    """Shape conflicts must be reported, not silently dropped (r1 verdict)."""
+    import warnings as w
+
+    cfg, model, opt = _tiny()
+    path = os.path.join(tmp_path, "c.pt")
+    save_checkpoint(path, model, opt, step=1)
+    cfg2 = GPTConfig(n_layer=2, n_head=2, n_embd=32, block_size=32,
+                     vocab_size=128)  # different vocab -> wte/lm_head clash
+    model2 = Single(GPT2Model(cfg2))
+    with w.catch_warnings(record=True) as rec:
+        w.simplefilter("always")
+        step, report = load_checkpoint(path, model2, return_report=True)
+    assert step == 1
+    assert report.mismatched, "vocab-size conflict must be surfaced"
+    names = [m[0] for m in report.mismatched]
+    assert any("wte" in n for n in names)
+    assert any("checkpoint" in str(r.message) for r in rec)
+    assert not report.ok()
+
+
+def test_rng_and_step_restored(tmp_path):
+    cfg, model, opt = _tiny()
+    _step(model, opt, 0)
+    torch.manual_seed(77)
+    torch.rand(3)  # advance the stream to a nontrivial state
+    path = os.path.join(tmp_path, "r.pt")
+    save_checkpoint(path, model, opt, step=5)
+    expected = torch.rand(4)  # what the stream yields after the save point
+    torch.manual_seed(1234)  # scramble
+    _, model2, opt2 = _tiny()
+    step, report = load_checkpoint(path, model2, opt2, return_report=True)
+    assert step == 5 and report.rng_restored and report.ok()
+    torch.testing.assert_close(torch.rand(4), expected)
+
+
+def test_zero2_reshard_world2_to_world3(tmp_path):
+    """ZeRO checkpoint saved at world 2 resumes at world 3: every rank
+    merges the two shard files and re-shards optimizer state under the new
+    partition; continuation losses match the uninterrupted world-2 run."""
+    import pytest
+
+    from tests.dist_utils import run_distributed
+    from tests import _dist_workers as w
+
+    path = os.path.join(tmp_path, "z2.pt")
+    ref = run_distributed(w.zero_ckpt_train_save, world=2,
+                          args=("zero2", path))
+    assert os.path.exists(path)
+    assert os.path.exists(path.replace(".pt", ".rank1.pt"))
+    resumed = run_distributed(w.zero_ckpt_resume, world=3,
+                              args=("zero2", path))
+    for r in range(3):
+        assert resumed[r] == pytest.approx(ref[0], rel=1e-4), (
+            resumed[r], ref[0])
+
+
 def test_checkpoint_model_only(tmp_path):
     cfg, model, opt = _tiny()
     _step(model, opt, 0)
