@@ -144,5 +144,8 @@ def _train(wrapper_name, force, steps=4):
 def test_forced_rccl_training_matches_unforced(strategy):
     forced = _train(strategy, force=True)
     plain = _train(strategy, force=False)
-    assert forced == plain, (forced, plain)
+    # not bit-exact: embedding-bwd scatter and the CE loss accumulator use
+    # fp32 atomics whose order varies run to run (~1e-6 relative, measured);
+    # comm-stream lifetime corruption would diverge far beyond this
+    assert forced == pytest.approx(plain, rel=1e-4), (forced, plain)
     assert forced[-1] < forced[0]
