@@ -34,6 +34,9 @@ class GPTConfig:
     # "fused": CDNA4 flash-style kernel (composite torch on CPU);
     # "math": composite torch attention everywhere.
     attention: str = "fused"
+    # row-chunked fused lm_head projection + cross-entropy (loss-only path
+    # returns logits=None; inference with targets=None is unaffected)
+    fused_lm_head: bool = True
 
     @classmethod
     def gpt2_small(cls, **kw):
@@ -185,6 +188,13 @@ class GPT2Model(nn.Module):
             _, y = ln_f.forward_fused(x, delta)
         else:
             y = ln_f(x + delta)
+        if (targets is not None and self.config.fused_lm_head
+                and hasattr(self.lm_head, "project_cross_entropy")):
+            # fused row-chunked projection+loss: full (B*T, V) logits are
+            # never materialized (the reference holds the 3.3 GB tensor for
+            # backward, example/model.py:152-156). Training callers read
+            # only the loss; logits are None on this path by design.
+            return None, self.lm_head.project_cross_entropy(y, targets)
         logits = self.lm_head(y)
         loss = None
         if targets is not None:

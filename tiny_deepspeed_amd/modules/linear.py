@@ -12,10 +12,92 @@ Backward contract: `backward_callback(dy, x, weight)` returns
 with no communication.
 """
 
+import os
+
 import torch
 import torch.nn as nn
 
 from .. import ops
+
+
+def _ce_chunk_rows(V):
+    """Row-chunk size for the fused projection+loss: the largest chunk whose
+    bf16 logits block (~C*V*2 bytes) stays within the MI355X LLC (256 MB
+    Infinity Cache) with headroom, so the GEMM-written logits are consumed
+    by the CE kernel without a round trip through HBM3E."""
+    env = os.environ.get("TDSA_CE_CHUNK")
+    if env:
+        return int(env)
+    c = (96 << 20) // max(V, 1)
+    return max(1024, (c // 256) * 256)
+
+
+class _LinearCEFn(torch.autograd.Function):
+    """Fused lm_head projection + softmax cross-entropy, row-chunked.
+
+    The reference materializes full (B*T, V) logits and runs F.cross_entropy
+    on them (/root/reference/example/model.py:152-156): at b32/gpt2-medium
+    that is a 3.3 GB tensor held for backward plus ~13 GB of HBM traffic
+    through the fwd write / CE read / dlogits write / dX+dW reads. Here the
+    projection GEMM and the CE kernels run per row-chunk (logits chunk stays
+    LLC-resident) and backward RECOMPUTES each chunk's logits from (x, lse),
+    so only the per-row logsumexp (4 bytes/row) survives the forward.
+
+    Weight handling goes through three overridable module hooks so every
+    parallel strategy keeps its semantics (same callback-factory design as
+    _LinearFn): ``_ce_weight_fwd``/``_ce_weight_bwd`` provide the (possibly
+    JIT-gathered, ZeRO-3) full weight; ``publish_weight_grad`` routes dW
+    into the strategy's collective (DDP all-reduce / ZeRO reduce-to-owner)
+    and returns None, or returns dW for plain autograd accumulation.
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, targets, module, ignore_index):
+        x2 = x.reshape(-1, x.shape[-1])
+        tg = targets.reshape(-1)
+        w = module._ce_weight_fwd()
+        R = x2.shape[0]
+        V = w.shape[0]
+        C = _ce_chunk_rows(V)
+        lse = torch.empty(R, dtype=torch.float32, device=x.device)
+        loss_sum = None
+        n_valid = None
+        for s in range(0, R, C):
+            e = min(s + C, R)
+            logits_c = ops.linear_forward(x2[s:e], w)
+            ls, lse_c, nv = ops.cross_entropy_fwd(logits_c, tg[s:e],
+                                                  ignore_index,
+                                                  tuner=module.tuner)
+            lse[s:e] = lse_c
+            loss_sum = ls if loss_sum is None else loss_sum + ls
+            n_valid = nv if n_valid is None else n_valid + nv
+        ctx.save_for_backward(x2, tg, lse, n_valid)
+        ctx.module = module
+        ctx.ignore_index = ignore_index
+        ctx.x_shape = x.shape
+        return loss_sum / n_valid.clamp(min=1)
+
+    @staticmethod
+    def backward(ctx, dloss):
+        x2, tg, lse, n_valid = ctx.saved_tensors
+        module = ctx.module
+        w = module._ce_weight_bwd()
+        R, E = x2.shape
+        V = w.shape[0]
+        C = _ce_chunk_rows(V)
+        nv = int(n_valid)
+        dx2 = torch.empty_like(x2)
+        dw32 = torch.zeros(V, E, dtype=torch.float32, device=x2.device)
+        for s in range(0, R, C):
+            e = min(s + C, R)
+            logits_c = ops.linear_forward(x2[s:e], w)
+            dl_c = ops.cross_entropy_bwd(dloss, logits_c, tg[s:e], lse[s:e],
+                                         nv, ctx.ignore_index,
+                                         tuner=module.tuner)
+            dx2[s:e] = ops.linear_input_grad(dl_c, w)
+            dw32 += ops.linear_weight_grad(dl_c, x2[s:e]).float()
+        dw = module.publish_weight_grad(dw32.to(w.dtype))
+        return dx2.view(ctx.x_shape), dw, None, None, None
 
 
 class _LinearFn(torch.autograd.Function):
@@ -64,6 +146,25 @@ class Linear(nn.Linear):
             )
         if db is not None and self.bias is not None:
             assert db.shape == self.bias.shape
+
+    # --- fused projection + cross-entropy hooks (see _LinearCEFn) --------
+    def _ce_weight_fwd(self):
+        return self.weight
+
+    def _ce_weight_bwd(self):
+        return self.weight
+
+    def publish_weight_grad(self, dw):
+        """No strategy: hand dW back to autograd for plain accumulation."""
+        return dw
+
+    def project_cross_entropy(self, x, targets, ignore_index=-100):
+        """loss = cross_entropy(x @ W.T, targets), row-chunked so the full
+        logits tensor is never materialized or saved. bias-free only (the
+        GPT-2 lm_head; reference example/model.py:135)."""
+        if self.bias is not None:
+            raise NotImplementedError("fused lm_head+CE requires bias=False")
+        return _LinearCEFn.apply(x, self.weight, targets, self, ignore_index)
 
     def forward(self, x):
         return _LinearFn.apply(x, self.weight, self.bias, self)

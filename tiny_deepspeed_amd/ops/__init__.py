@@ -17,7 +17,7 @@ from .layernorm import (layernorm_fwd, layernorm_fwd_res, layernorm_dx,
 from .embedding import embedding_forward, embedding_weight_grad
 from .gelu import gelu, gelu_fwd, gelu_bwd
 from .attention import causal_attention, fused_causal_attention
-from .cross_entropy import cross_entropy
+from .cross_entropy import cross_entropy, cross_entropy_fwd, cross_entropy_bwd
 from .optim_ops import adamw_step, sgd_step
 from .autotuner import RuntimeAutoTuner, default_tuner
 from ._ext import ext_available, get_ext
@@ -29,7 +29,7 @@ __all__ = [
     "embedding_forward", "embedding_weight_grad",
     "gelu", "gelu_fwd", "gelu_bwd",
     "causal_attention", "fused_causal_attention",
-    "cross_entropy",
+    "cross_entropy", "cross_entropy_fwd", "cross_entropy_bwd",
     "adamw_step", "sgd_step",
     "RuntimeAutoTuner", "default_tuner",
     "ext_available", "get_ext",

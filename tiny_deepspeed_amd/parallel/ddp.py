@@ -32,6 +32,12 @@ class Linear(base.Linear):
         dx = ops.linear_input_grad(dy, self.weight, tuner=self.tuner)
         return dx, None, None
 
+    def publish_weight_grad(self, dw):
+        """Fused lm_head+CE path: route dW into this strategy's collective
+        (all-reduce / reduce-to-owner per _mode) instead of autograd."""
+        publish_grad(self._comm, self.weight, dw, self._mode)
+        return None
+
 
 class LayerNorm(base.LayerNorm):
     _mode = ALLREDUCE

@@ -86,6 +86,18 @@ class _GatherMixin:
 
 
 class Linear(_GatherMixin, base.Linear):
+    # fused lm_head+CE hooks: consume the JIT-gathered full weight in both
+    # directions; dW is averaged-reduced to the owner and dropped elsewhere
+    def _ce_weight_fwd(self):
+        return self._take_bufs("_tdsa_next")["weight"]
+
+    def _ce_weight_bwd(self):
+        return self._take_bufs("_tdsa_prev")["weight"]
+
+    def publish_weight_grad(self, dw):
+        publish_grad(self._comm, self.weight, dw, REDUCE_SHARD)
+        return None
+
     def forward_callback(self, x, weight, bias):
         bufs = self._take_bufs("_tdsa_next")
         return ops.linear_forward(x, bufs["weight"], bufs["bias"],
