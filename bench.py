@@ -43,6 +43,7 @@ from tiny_deepspeed_amd import (
     Zero1, Zero1AdamW,
     Zero2, Zero2AdamW,
     Zero3, Zero3AdamW,
+    Zero2Flat, Zero2FlatAdamW,
     partition_tensors,
 )
 
@@ -52,6 +53,7 @@ WRAPPERS = {
     "zero1": (Zero1, Zero1AdamW),
     "zero2": (Zero2, Zero2AdamW),
     "zero3": (Zero3, Zero3AdamW),
+    "zero2flat": (Zero2Flat, Zero2FlatAdamW),
 }
 
 
@@ -98,7 +100,10 @@ def main():
     model = GPT2Model(config).to(device=device, dtype=dtype)
 
     wrapper_cls, optim_cls = WRAPPERS[args.parallel]
-    if args.parallel in ("zero1", "zero2", "zero3"):
+    if args.parallel == "zero2flat":
+        wrapped = wrapper_cls(model)
+        optimizer = optim_cls(wrapped, lr=1e-5, weight_decay=0.1)
+    elif args.parallel in ("zero1", "zero2", "zero3"):
         ranks_map = [f"cuda:{i}" if have_gpu else "cpu"
                      for i in range(world_size)]
         with torch.device("meta"):

@@ -182,6 +182,83 @@ def grad_accumulation(rank, world):
     return lin.weight.grad.clone()
 
 
+def zero2flat_losses(rank, world, bucket_bytes=32 << 20):
+    """Flat-bucket ZeRO-2: returns (losses, collective counts, n_buckets,
+    n_params) so the test can assert loss parity AND the O(#params) ->
+    O(#buckets) collective reduction."""
+    import tiny_deepspeed_amd as tdsa
+    from tiny_deepspeed_amd.models import GPT2Model
+
+    torch.manual_seed(0)
+    model = tdsa.Zero2Flat(GPT2Model(make_cfg()), bucket_bytes=bucket_bytes)
+    opt = tdsa.Zero2FlatAdamW(model, lr=1e-3, weight_decay=0.01)
+    comm = model.comm
+    counts = {"rs": 0, "ag": 0}
+    orig_rs = comm.reduce_scatter_avg
+    orig_ag = comm.all_gather_flat
+
+    def rs(out, inp):
+        counts["rs"] += 1
+        return orig_rs(out, inp)
+
+    def ag(out, inp):
+        counts["ag"] += 1
+        return orig_ag(out, inp)
+
+    comm.reduce_scatter_avg = rs
+    comm.all_gather_flat = ag
+    x, y = batch()
+    losses = []
+    for _ in range(ITERS):
+        model.require_backward_grad_sync = True
+        _, loss = model(x, y)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    n_params = len(list(model.named_parameters()))
+    return losses, counts, len(model.engine.buckets), n_params
+
+
+def zero2flat_grad_accum(rank, world):
+    """Two no-sync microbatches then an armed one: the flat slots must
+    accumulate; compare against a single armed pass over summed grads via
+    the resulting loss trajectory (parity with per-tensor zero2)."""
+    import tiny_deepspeed_amd as tdsa
+    from tiny_deepspeed_amd.models import GPT2Model
+
+    def run(flat):
+        torch.manual_seed(0)
+        m = GPT2Model(make_cfg())
+        if flat:
+            model = tdsa.Zero2Flat(m)
+            opt = tdsa.Zero2FlatAdamW(model, lr=1e-3, weight_decay=0.01)
+        else:
+            parts, _ = tdsa.partition_tensors(
+                OrderedDict((n, p) for n, p in m.named_parameters()),
+                ["cpu"] * world)
+            model = tdsa.Zero2(m, parts)
+            opt = tdsa.Zero2AdamW(model.named_parameters(), lr=1e-3,
+                                  weight_decay=0.01,
+                                  param_part_table=parts,
+                                  ranks_map=["cpu"] * world)
+        losses = []
+        for it in range(2):
+            for micro in range(3):
+                g = torch.Generator().manual_seed(10 * it + micro)
+                x = torch.randint(0, CFG["vocab_size"],
+                                  (2, CFG["block_size"]), generator=g)
+                y = torch.randint(0, CFG["vocab_size"],
+                                  (2, CFG["block_size"]), generator=g)
+                model.require_backward_grad_sync = micro == 2
+                _, loss = model(x, y)
+                loss.backward()
+                losses.append(loss.item())
+            opt.step()
+        return losses
+
+    return run(True), run(False)
+
+
 def _build_zero(rank, world, strategy):
     import tiny_deepspeed_amd as tdsa
     from tiny_deepspeed_amd.models import GPT2Model

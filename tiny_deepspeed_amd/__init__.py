@@ -18,6 +18,7 @@ from .parallel import (
     Zero1, Zero1SGD, Zero1AdamW,
     Zero2, Zero2SGD, Zero2AdamW,
     Zero3, Zero3SGD, Zero3AdamW,
+    Zero2Flat, Zero2FlatSGD, Zero2FlatAdamW,
     Single,
     partition_tensors,
 )
@@ -30,6 +31,7 @@ __all__ = [
     "Zero1", "Zero1SGD", "Zero1AdamW",
     "Zero2", "Zero2SGD", "Zero2AdamW",
     "Zero3", "Zero3SGD", "Zero3AdamW",
+    "Zero2Flat", "Zero2FlatSGD", "Zero2FlatAdamW",
     "Single",
     "partition_tensors",
 ]

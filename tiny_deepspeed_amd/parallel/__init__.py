@@ -12,6 +12,7 @@ from .ddp import DDP, DDPSGD, DDPAdamW
 from .zero1 import Zero1, Zero1SGD, Zero1AdamW
 from .zero2 import Zero2, Zero2SGD, Zero2AdamW
 from .zero3 import Zero3, Zero3SGD, Zero3AdamW
+from .flat import Zero2Flat, Zero2FlatSGD, Zero2FlatAdamW
 
 __all__ = [
     "CommContext", "default_comm",
@@ -21,4 +22,5 @@ __all__ = [
     "Zero1", "Zero1SGD", "Zero1AdamW",
     "Zero2", "Zero2SGD", "Zero2AdamW",
     "Zero3", "Zero3SGD", "Zero3AdamW",
+    "Zero2Flat", "Zero2FlatSGD", "Zero2FlatAdamW",
 ]

@@ -17,12 +17,18 @@ from .wrapper import take_sync
 ALLREDUCE = "allreduce"        # DDP: average everywhere, grad replicated
 REDUCE_KEEP = "reduce_keep"    # ZeRO-1: average onto owner, local grad kept
 REDUCE_SHARD = "reduce_shard"  # ZeRO-2/3: average onto owner, dropped elsewhere
+FLAT = "flat"                  # flat-bucket ZeRO-2: slot into the engine's
+#                                bucket; one reduce_scatter per full bucket
 
 
 def publish_grad(comm, param, dw, mode):
     """Accumulate dw into param.grad, launching the strategy's collective
     (async, comm stream) when the once-per-iteration latch is armed."""
     if dw is None:
+        return
+    if mode == FLAT:
+        # the engine owns accumulation, bucket completion and the collective
+        param._tdsa_engine.publish(param, dw, take_sync(param))
         return
     # ZeRO-3 non-owners have 0-numel param storage: full-shape gradients
     # accumulate in a side slot instead of .grad (torch enforces shape).

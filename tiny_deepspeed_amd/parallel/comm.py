@@ -236,6 +236,60 @@ class CommContext:
                                   async_op=True)
             self._pending_flat[channel].append((work, flat, bucket, numels, src))
 
+    def reduce_scatter_avg(self, out_shard, in_flat):
+        """Average-reduce `in_flat` (numel == world * shard) across ranks and
+        scatter equal shards: this rank's shard lands in `out_shard`.
+        Async on the REDUCE stream. The flat-bucket ZeRO-2 grad collective —
+        one call replaces a bucketful of per-tensor reduces."""
+        if self._inactive():
+            out_shard.copy_(in_flat[: out_shard.numel()])
+            return out_shard
+
+        def run(async_op=False):
+            in_flat.div_(self.world_size)
+            if dist.get_backend(self.pg[REDUCE]) == "gloo":
+                # gloo lacks reduce_scatter_tensor: all-reduce then slice
+                # (CPU test path only; synchronous is fine there)
+                work = dist.all_reduce(in_flat, op=dist.ReduceOp.SUM,
+                                       group=self.pg[REDUCE],
+                                       async_op=async_op)
+                if work is not None:
+                    work.wait()
+                n = out_shard.numel()
+                out_shard.copy_(in_flat[self.rank * n:(self.rank + 1) * n])
+                return None
+            return dist.reduce_scatter_tensor(
+                out_shard, in_flat, op=dist.ReduceOp.SUM,
+                group=self.pg[REDUCE], async_op=async_op)
+
+        self._launch([out_shard, in_flat], run, REDUCE)
+        return out_shard
+
+    def all_gather_flat(self, out_flat, in_shard):
+        """All-gather equal shards into `out_flat` (in-place friendly:
+        in_shard may alias out_flat's own-rank slice). Async, REDUCE stream."""
+        if self._inactive():
+            n = in_shard.numel()
+            if out_flat[:n].data_ptr() != in_shard.data_ptr():
+                out_flat[:n].copy_(in_shard)
+            return out_flat
+
+        def run(async_op=False):
+            if dist.get_backend(self.pg[REDUCE]) == "gloo":
+                chunks = list(out_flat.chunk(self.world_size))
+                work = dist.all_gather(chunks, in_shard.clone(),
+                                       group=self.pg[REDUCE],
+                                       async_op=async_op)
+                if work is not None:
+                    work.wait()
+                return None
+            return dist.all_gather_into_tensor(out_flat, in_shard,
+                                               group=self.pg[REDUCE],
+                                               async_op=async_op)
+
+        self._launch([out_flat, in_shard], run, REDUCE)
+        return out_flat
+
     def all_reduce_scalar_avg(self, t):
         """Synchronous scalar average (loss logging)."""
         if self._inactive():
