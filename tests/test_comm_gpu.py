@@ -108,6 +108,10 @@ def _train(wrapper_name, force, steps=4):
         if wrapper_name == "single":
             wrapped = Single(model)
             opt = AdamW(wrapped.named_parameters(), lr=1e-3)
+        elif wrapper_name == "zero2flat":
+            from tiny_deepspeed_amd import Zero2Flat, Zero2FlatAdamW
+            wrapped = Zero2Flat(model, comm=comm, bucket_bytes=1 << 20)
+            opt = Zero2FlatAdamW(wrapped, lr=1e-3)
         elif wrapper_name == "ddp":
             wrapped = DDP(model, comm=comm)
             opt = DDPAdamW(wrapped.named_parameters(), lr=1e-3, comm=comm)
@@ -140,7 +144,8 @@ def _train(wrapper_name, force, steps=4):
         os.environ.pop("TDSA_AUTOTUNE", None)
 
 
-@pytest.mark.parametrize("strategy", ["ddp", "zero1", "zero2", "zero3"])
+@pytest.mark.parametrize("strategy", ["ddp", "zero1", "zero2", "zero3",
+                                      "zero2flat"])
 def test_forced_rccl_training_matches_unforced(strategy):
     forced = _train(strategy, force=True)
     plain = _train(strategy, force=False)
