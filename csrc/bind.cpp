@@ -241,11 +241,21 @@ std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets,
 
 at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets,
                              at::Tensor lse, double dloss, int64_t n_valid,
-                             int64_t ignore_index) {
+                             int64_t ignore_index,
+                             c10::optional<at::Tensor> out) {
   CHECK_IN(logits); CHECK_IN(targets); CHECK_IN(lse);
   const long long R = logits.size(0);
   const int V = logits.size(1);
-  auto dlogits = at::empty_like(logits);
+  at::Tensor dlogits;
+  if (out.has_value()) {  // e.g. a row-slice of the fused-CE dlogits buffer
+    dlogits = *out;
+    CHECK_IN(dlogits);
+    TORCH_CHECK(dlogits.sizes() == logits.sizes()
+                && dlogits.scalar_type() == logits.scalar_type(),
+                "cross_entropy_bwd: out must match logits");
+  } else {
+    dlogits = at::empty_like(logits);
+  }
   const float scale = (float)(dloss / (double)std::max<int64_t>(n_valid, 1));
   check_hip(tdsa_ce_bwd(logits.data_ptr(), (const long long*)targets.data_ptr<int64_t>(),
                         lse.data_ptr<float>(), dlogits.data_ptr(), R, V, scale,
@@ -512,7 +522,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("embedding_fwd", &embedding_fwd);
   mod.def("embedding_bwd", &embedding_bwd);
   mod.def("cross_entropy_fwd", &cross_entropy_fwd);
-  mod.def("cross_entropy_bwd", &cross_entropy_bwd);
+  mod.def("cross_entropy_bwd", &cross_entropy_bwd, py::arg("logits"),
+          py::arg("targets"), py::arg("lse"), py::arg("dloss"),
+          py::arg("n_valid"), py::arg("ignore_index"),
+          py::arg("out") = py::none());
   mod.def("adamw_step", &adamw_step);
   mod.def("adamw_step_multi", &adamw_step_multi);
   mod.def("sgd_step", &sgd_step);

@@ -24,12 +24,13 @@ def _ce_chunk_rows(V):
     """Row-chunk size for the fused projection+loss: the largest chunk whose
     bf16 logits block (~C*V*2 bytes) stays within the MI355X LLC (256 MB
     Infinity Cache) with headroom, so the GEMM-written logits are consumed
-    by the CE kernel without a round trip through HBM3E."""
+    by the CE kernel without a round trip through HBM3E. Rounded UP to 256
+    so the chunk GEMMs keep full wave efficiency."""
     env = os.environ.get("TDSA_CE_CHUNK")
     if env:
         return int(env)
-    c = (96 << 20) // max(V, 1)
-    return max(1024, (c // 256) * 256)
+    c = (112 << 20) // max(V, 1)
+    return max(1024, ((c + 255) // 256) * 256)
 
 
 class _LinearCEFn(torch.autograd.Function):
@@ -86,17 +87,20 @@ class _LinearCEFn(torch.autograd.Function):
         V = w.shape[0]
         C = _ce_chunk_rows(V)
         nv = int(n_valid)
-        dx2 = torch.empty_like(x2)
-        dw32 = torch.zeros(V, E, dtype=torch.float32, device=x2.device)
+        # recompute logits per chunk, but materialize the FULL bf16 dlogits
+        # (transient, freed at return) so dX and dW run as single
+        # full-reduction GEMMs — a first per-chunk-accumulation version paid
+        # ~10 ms/step in fp32 dW add/cast elementwise traffic (profiled)
+        dlogits = torch.empty(R, V, dtype=x2.dtype, device=x2.device)
         for s in range(0, R, C):
             e = min(s + C, R)
             logits_c = ops.linear_forward(x2[s:e], w)
-            dl_c = ops.cross_entropy_bwd(dloss, logits_c, tg[s:e], lse[s:e],
-                                         nv, ctx.ignore_index,
-                                         tuner=module.tuner)
-            dx2[s:e] = ops.linear_input_grad(dl_c, w)
-            dw32 += ops.linear_weight_grad(dl_c, x2[s:e]).float()
-        dw = module.publish_weight_grad(dw32.to(w.dtype))
+            ops.cross_entropy_bwd(
+                dloss, logits_c, tg[s:e], lse[s:e], nv, ctx.ignore_index,
+                tuner=module.tuner, out=dlogits[s:e])
+        dx2 = ops.linear_input_grad(dlogits, w)
+        dw = module.publish_weight_grad(
+            ops.linear_weight_grad(dlogits, x2))
         return dx2.view(ctx.x_shape), dw, None, None, None
 
 

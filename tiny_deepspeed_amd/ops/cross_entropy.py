@@ -51,14 +51,14 @@ def cross_entropy_fwd(logits2d, targets, ignore_index=-100, tuner=None):
     return ce_fwd_hip(logits2d, targets, ignore_index)
 
 
-def ce_bwd_hip(dloss, logits2d, targets, lse, n_valid, ignore_index):
+def ce_bwd_hip(dloss, logits2d, targets, lse, n_valid, ignore_index, out=None):
     return _ext.get_ext().cross_entropy_bwd(
         logits2d.contiguous(), targets.contiguous(), lse,
-        float(dloss), int(n_valid), ignore_index,
+        float(dloss), int(n_valid), ignore_index, out,
     )
 
 
-def ce_bwd_torch(dloss, logits2d, targets, lse, n_valid, ignore_index):
+def ce_bwd_torch(dloss, logits2d, targets, lse, n_valid, ignore_index, out=None):
     lf = logits2d.float()
     soft = (lf - lse.unsqueeze(-1)).exp()
     valid = (targets != ignore_index).unsqueeze(1)
@@ -68,17 +68,26 @@ def ce_bwd_torch(dloss, logits2d, targets, lse, n_valid, ignore_index):
     )
     scale = float(dloss) / max(int(n_valid), 1)
     dlogits = torch.where(valid, soft * scale, torch.zeros_like(soft))
+    if out is not None:
+        out.copy_(dlogits.to(logits2d.dtype))
+        return out
     return dlogits.to(logits2d.dtype)
 
 
-def cross_entropy_bwd(dloss, logits2d, targets, lse, n_valid, ignore_index=-100, tuner=None):
+def cross_entropy_bwd(dloss, logits2d, targets, lse, n_valid,
+                      ignore_index=-100, tuner=None, out=None):
+    """out: optional pre-allocated dlogits destination (e.g. a row-slice of
+    the fused lm_head+CE full-dlogits buffer) — skips an extra copy."""
     if not (_ext.use_native(logits2d) and _kernel_supported(logits2d)):
-        return ce_bwd_torch(dloss, logits2d, targets, lse, n_valid, ignore_index)
+        return ce_bwd_torch(dloss, logits2d, targets, lse, n_valid,
+                            ignore_index, out)
     tuner = tuner if tuner is not None else default_tuner()
     if tuner is not None:
         return tuner.choose("ce_bwd", [ce_bwd_hip, ce_bwd_torch],
-                            dloss, logits2d, targets, lse, n_valid, ignore_index)
-    return ce_bwd_hip(dloss, logits2d, targets, lse, n_valid, ignore_index)
+                            dloss, logits2d, targets, lse, n_valid,
+                            ignore_index, out)
+    return ce_bwd_hip(dloss, logits2d, targets, lse, n_valid, ignore_index,
+                      out)
 
 
 class _CrossEntropyFn(torch.autograd.Function):
