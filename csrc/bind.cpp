@@ -55,6 +55,9 @@ hipError_t tdsa_attn_bwd(const void*, const void*, const void*, const void*,
                          const float*, const void*, void*, void*, void*, float*,
                          long long, long long, int, float, const long long*,
                          const long long*, const long long*, hipStream_t);
+int tdsa_gemm_tn_splits(long long M, int N, int K);
+hipError_t tdsa_gemm_tn(const void*, const void*, float*, long long, int, int,
+                        hipStream_t);
 hipError_t tdsa_dbg_mfma(const void*, const void*, float*, int, hipStream_t);
 hipError_t tdsa_dbg_mfma32(const void*, const void*, float*, hipStream_t);
 hipError_t tdsa_dbg_stage(const void*, void*, int, hipStream_t);
@@ -395,6 +398,27 @@ void sgd_step(at::Tensor param, at::Tensor grad, at::Tensor buf,
             "sgd_step");
 }
 
+// ---- TN GEMM (linear dW autotuner candidate) ------------------------------
+at::Tensor gemm_tn(at::Tensor dy2, at::Tensor x2) {
+  CHECK_IN(dy2); CHECK_IN(x2);
+  TORCH_CHECK(dy2.dim() == 2 && x2.dim() == 2, "gemm_tn expects 2-D inputs");
+  TORCH_CHECK(dy2.scalar_type() == at::kBFloat16
+              && x2.scalar_type() == at::kBFloat16, "gemm_tn is bf16-only");
+  TORCH_CHECK(dy2.size(0) == x2.size(0), "reduction dims differ");
+  const long long M = dy2.size(0);
+  const int N = dy2.size(1);
+  const int K = x2.size(1);
+  const int splits = tdsa_gemm_tn_splits(M, N, K);
+  TORCH_CHECK(splits > 0, "gemm_tn: unsupported shape (need M%64==0, "
+              "N%128==0, K%128==0), got ", M, "x", N, "/", K);
+  auto f32 = dy2.options().dtype(at::kFloat);
+  auto dw32 = splits > 1 ? at::zeros({N, K}, f32) : at::empty({N, K}, f32);
+  check_hip(tdsa_gemm_tn(dy2.data_ptr(), x2.data_ptr(),
+                         dw32.data_ptr<float>(), M, N, K, cur_stream()),
+            "gemm_tn");
+  return dw32.to(at::kBFloat16);
+}
+
 // ---- attention ------------------------------------------------------------
 static void check_attn_tensor(const at::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
@@ -526,6 +550,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           py::arg("targets"), py::arg("lse"), py::arg("dloss"),
           py::arg("n_valid"), py::arg("ignore_index"),
           py::arg("out") = py::none());
+  mod.def("gemm_tn", &gemm_tn);
   mod.def("adamw_step", &adamw_step);
   mod.def("adamw_step_multi", &adamw_step_multi);
   mod.def("sgd_step", &sgd_step);

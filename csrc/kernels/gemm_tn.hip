@@ -1,0 +1,166 @@
+// Hand-written CDNA4 TN GEMM for the Linear weight gradient:
+//
+//   dw[N,K] = dy[M,N]^T @ x[M,K]     (bf16 in, fp32 accumulate)
+//
+// The ONE GEMM shape family where the reduction dim (M = B*T, 32k typical)
+// is the ROW dim of BOTH row-major operands. That makes both MFMA operands
+// reachable through the same ds_read_tr16_b64 hardware-transpose read of
+// row-major staged [64][64] LDS images (mfma.h: the A and B fragment
+// layouts of v_mfma_f32_32x32x16_bf16 share one lane->(k-elem, 32-axis)
+// map) — no transposed staging, no bank conflicts, no strided global reads.
+// hipBLASLt's weakest library shape per profiles/gemm_shapes_hipblaslt.txt;
+// registered as the autotuner's first candidate for linear_dw
+// (ops/linear.py), so the library wins any shape where it is faster.
+//
+// Geometry: 4 waves / workgroup, 128(N) x 128(K) output tile; wave w owns
+// the 32-row n-band [32w, 32w+32) x 128 k. Per 64-row m-chunk each wave
+// issues 4 A tr-reads + 16 B tr-reads + 16 MFMAs from four 16 KB swizzled
+// LDS images (dy 64x128, x 64x128), with the next chunk's global loads
+// issued before the compute (fetch/put split — they fly under the MFMAs).
+// The output is tiny (N*K vs M*K inputs), so workgroups additionally
+// split the M reduction (grid.y) and combine 128x128 fp32 partials with
+// atomicAdd into a zeroed fp32 buffer (skipped when grid.y == 1); the
+// caller converts to bf16.
+//
+// Reference hot path being replaced:
+// /root/reference/tiny_deepspeed/core/module/ops/linear.py:59-68.
+#include "common.h"
+#include "mfma.h"
+
+#include <cstdlib>
+
+namespace tdsa {
+
+template <bool ATOMIC>
+__launch_bounds__(256)
+__global__ void gemm_tn_kernel(const bf16* __restrict__ dy,
+                               const bf16* __restrict__ x,
+                               float* __restrict__ dw, long long M, int N,
+                               int K, long long m_per_split) {
+  const int tiles_k = K >> 7;
+  const int tn = blockIdx.x / tiles_k;
+  const int tk = blockIdx.x % tiles_k;
+  const int n0 = tn << 7;
+  const int k0 = tk << 7;
+  const long long m_begin = (long long)blockIdx.y * m_per_split;
+  long long m_end = m_begin + m_per_split;
+  if (m_end > M) m_end = M;
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wave = tid >> 6;
+  const int h32 = lane >> 5;
+
+  __shared__ char lds[4 * 64 * 128];  // dy lo/hi, x lo/hi images (16 KB ea)
+  constexpr int IMG = 64 * 128;
+
+  Stage<256> st[4] = {
+      Stage<256>(dy + m_begin * N + n0, tid, N),
+      Stage<256>(dy + m_begin * N + n0 + 64, tid, N),
+      Stage<256>(x + m_begin * K + k0, tid, K),
+      Stage<256>(x + m_begin * K + k0 + 64, tid, K),
+  };
+
+  int trb[2][2];
+  tr_bases(lane, trb);
+  const char* a_img = lds + (wave >> 1) * IMG;
+  const int dta = wave & 1;
+
+  f32x16 acc[4];
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) acc[dt][r] = 0.f;
+
+  const int nchunks = m_end > m_begin ? (int)((m_end - m_begin) >> 6) : 0;
+  if (nchunks > 0) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      st[i].run(lds + i * IMG);
+      st[i].advance();
+    }
+  }
+  for (int c = 0; c < nchunks; ++c) {
+    __syncthreads();
+    short8v pend[4][Stage<256>::REPS];
+    const bool more = c + 1 < nchunks;
+    if (more) {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) st[i].fetch(pend[i]);
+    }
+#pragma unroll
+    for (int S = 0; S < 4; ++S) {
+      const bfrag a = tr_bfrag(a_img, trb[dta][0] + S * 2048,
+                               trb[dta][1] + S * 2048);
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        const bfrag b = tr_bfrag(lds + (2 + (dt >> 1)) * IMG,
+                                 trb[dt & 1][0] + S * 2048,
+                                 trb[dt & 1][1] + S * 2048);
+        acc[dt] = MFMA32(a, b, acc[dt]);
+      }
+    }
+    __syncthreads();
+    if (more) {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        st[i].put(lds + i * IMG, pend[i]);
+        st[i].advance();
+      }
+    }
+  }
+
+  // epilogue: C[32x32] lane l reg r -> row crow(r,h32), col l%32
+  const int col = k0 + (lane & 31);
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const long long idx =
+          (long long)(n0 + wave * 32 + crow(r, h32)) * K + col + dt * 32;
+      if (ATOMIC)
+        atomicAdd(dw + idx, acc[dt][r]);
+      else
+        dw[idx] = acc[dt][r];
+    }
+  }
+}
+
+}  // namespace tdsa
+
+using namespace tdsa;
+
+extern "C" {
+
+// Returns the split count the launcher will use (for the caller to decide
+// whether the fp32 buffer needs zeroing); <=0 means shape unsupported.
+int tdsa_gemm_tn_splits(long long M, int N, int K) {
+  if (M <= 0 || (M & 63) || (N & 127) || (K & 127)) return 0;
+  const long long tiles = (long long)(N >> 7) * (K >> 7);
+  // enough workgroups to fill 256 CUs several times over
+  long long want = (2048 + tiles - 1) / tiles;
+  long long maxs = M >> 6;  // at least one 64-chunk per split
+  if (want > maxs) want = maxs;
+  if (const char* v = getenv("TDSA_GEMM_TN_SPLITS")) want = atoll(v);
+  if (want < 1) want = 1;
+  return (int)want;
+}
+
+hipError_t tdsa_gemm_tn(const void* dy, const void* x, float* dw, long long M,
+                        int N, int K, hipStream_t stream) {
+  const int splits = tdsa_gemm_tn_splits(M, N, K);
+  if (splits <= 0) return hipErrorInvalidValue;
+  const long long chunks_per_split = ((M >> 6) + splits - 1) / splits;
+  const long long m_per_split = chunks_per_split << 6;
+  dim3 grid((N >> 7) * (K >> 7), splits);
+  if (splits == 1)
+    hipLaunchKernelGGL((gemm_tn_kernel<false>), grid, dim3(256), 0, stream,
+                       (const bf16*)dy, (const bf16*)x, dw, M, N, K,
+                       m_per_split);
+  else
+    hipLaunchKernelGGL((gemm_tn_kernel<true>), grid, dim3(256), 0, stream,
+                       (const bf16*)dy, (const bf16*)x, dw, M, N, K,
+                       m_per_split);
+  return hipGetLastError();
+}
+
+}  // extern "C"

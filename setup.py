@@ -22,6 +22,7 @@ SOURCES = [
     "csrc/kernels/cross_entropy.hip",
     "csrc/kernels/optim.hip",
     "csrc/kernels/attention.hip",
+    "csrc/kernels/gemm_tn.hip",
     "csrc/kernels/debug.hip",
 ]
 
