@@ -67,6 +67,9 @@ def main():
     p.add_argument("--seq", type=int, default=1024)
     p.add_argument("--parallel", default="zero2",
                    choices=list(WRAPPERS.keys()))
+    p.add_argument("--fused-lmhead", type=int, default=1,
+                   help="1: row-chunked fused lm_head+CE (default); 0: "
+                        "materialized logits path (A/B comparisons)")
     args = p.parse_args()
 
     have_gpu = torch.cuda.is_available()
@@ -94,7 +97,8 @@ def main():
         args.seq = min(args.seq, 64)
 
     model_name = args.model
-    config = GPTConfig.named(model_name, block_size=max(args.seq, 64))
+    config = GPTConfig.named(model_name, block_size=max(args.seq, 64),
+                             fused_lm_head=bool(args.fused_lmhead))
 
     torch.manual_seed(1234)
     model = GPT2Model(config).to(device=device, dtype=dtype)
