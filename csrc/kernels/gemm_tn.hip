@@ -31,18 +31,41 @@
 
 namespace tdsa {
 
-template <bool ATOMIC>
-__launch_bounds__(256)
+// XCD-group remap: the dispatcher places linear block b on XCD b % 8 with
+// x varying fastest, so the tiles_k workgroups that share one (tn, split)
+// — and therefore stream the SAME dy column-slice — would land on 8
+// different XCDs' L2s and fetch it 8x from HBM. Remap so they share an
+// XCD. Bijective when tiles_n * splits % 8 == 0; identity otherwise.
+DEV_INLINE void tn_remap(int gx, int gy, int& tile, int& split, int tiles_k) {
+  const int groups = (gx / tiles_k) * gy;  // (tn, split) pairs
+  if (groups % 8) return;
+  const long long id = (long long)split * gx + tile;
+  const long long window = 8LL * tiles_k;
+  const int g = (int)((id / window) * 8 + (id % 8));
+  const int tk = (int)((id % window) / 8);
+  const int tn = g / gy;
+  split = g % gy;
+  tile = tn * tiles_k + tk;
+}
+
+// NW waves; output tile (NW*32) x 128. Wave w owns n-band [32w, 32w+32) x
+// 128 k. NW=8 halves the staged bytes per FLOP vs NW=4 (the dy slice is
+// amortized over twice the MFMA work) — used when N % 256 == 0.
+template <int NW, bool ATOMIC>
+__launch_bounds__(NW * WAVE)
 __global__ void gemm_tn_kernel(const bf16* __restrict__ dy,
                                const bf16* __restrict__ x,
                                float* __restrict__ dw, long long M, int N,
-                               int K, long long m_per_split) {
+                               int K, long long m_per_split, int remap) {
   const int tiles_k = K >> 7;
-  const int tn = blockIdx.x / tiles_k;
-  const int tk = blockIdx.x % tiles_k;
-  const int n0 = tn << 7;
+  int tile = blockIdx.x;
+  int split = blockIdx.y;
+  if (remap) tn_remap(gridDim.x, gridDim.y, tile, split, tiles_k);
+  const int tn = tile / tiles_k;
+  const int tk = tile % tiles_k;
+  const int n0 = tn * (NW * 32);
   const int k0 = tk << 7;
-  const long long m_begin = (long long)blockIdx.y * m_per_split;
+  const long long m_begin = (long long)split * m_per_split;
   long long m_end = m_begin + m_per_split;
   if (m_end > M) m_end = M;
   const int tid = threadIdx.x;
@@ -50,15 +73,18 @@ __global__ void gemm_tn_kernel(const bf16* __restrict__ dy,
   const int wave = tid >> 6;
   const int h32 = lane >> 5;
 
-  __shared__ char lds[4 * 64 * 128];  // dy lo/hi, x lo/hi images (16 KB ea)
-  constexpr int IMG = 64 * 128;
+  constexpr int NT = NW * WAVE;
+  constexpr int NIMG = NW / 2 + 2;         // dy images + 2 x images
+  constexpr int IMG = 64 * 128;            // one swizzled [64][64] tile, 8 KB
+  __shared__ char lds[NIMG * IMG];
 
-  Stage<256> st[4] = {
-      Stage<256>(dy + m_begin * N + n0, tid, N),
-      Stage<256>(dy + m_begin * N + n0 + 64, tid, N),
-      Stage<256>(x + m_begin * K + k0, tid, K),
-      Stage<256>(x + m_begin * K + k0 + 64, tid, K),
-  };
+  // images 0..NW/2-1: dy 64-col slices; images NW/2, NW/2+1: x lo/hi
+  Stage<NT> st[NIMG];
+#pragma unroll
+  for (int i = 0; i < NW / 2; ++i)
+    st[i].init(dy + m_begin * N + n0 + 64 * i, tid, N);
+  st[NW / 2].init(x + m_begin * K + k0, tid, K);
+  st[NW / 2 + 1].init(x + m_begin * K + k0 + 64, tid, K);
 
   int trb[2][2];
   tr_bases(lane, trb);
@@ -74,18 +100,18 @@ __global__ void gemm_tn_kernel(const bf16* __restrict__ dy,
   const int nchunks = m_end > m_begin ? (int)((m_end - m_begin) >> 6) : 0;
   if (nchunks > 0) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < NIMG; ++i) {
       st[i].run(lds + i * IMG);
       st[i].advance();
     }
   }
   for (int c = 0; c < nchunks; ++c) {
     __syncthreads();
-    short8v pend[4][Stage<256>::REPS];
+    short8v pend[NIMG][Stage<NT>::REPS];
     const bool more = c + 1 < nchunks;
     if (more) {
 #pragma unroll
-      for (int i = 0; i < 4; ++i) st[i].fetch(pend[i]);
+      for (int i = 0; i < NIMG; ++i) st[i].fetch(pend[i]);
     }
 #pragma unroll
     for (int S = 0; S < 4; ++S) {
@@ -93,7 +119,7 @@ __global__ void gemm_tn_kernel(const bf16* __restrict__ dy,
                                trb[dta][1] + S * 2048);
 #pragma unroll
       for (int dt = 0; dt < 4; ++dt) {
-        const bfrag b = tr_bfrag(lds + (2 + (dt >> 1)) * IMG,
+        const bfrag b = tr_bfrag(lds + (NW / 2 + (dt >> 1)) * IMG,
                                  trb[dt & 1][0] + S * 2048,
                                  trb[dt & 1][1] + S * 2048);
         acc[dt] = MFMA32(a, b, acc[dt]);
@@ -102,7 +128,7 @@ __global__ void gemm_tn_kernel(const bf16* __restrict__ dy,
     __syncthreads();
     if (more) {
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
+      for (int i = 0; i < NIMG; ++i) {
         st[i].put(lds + i * IMG, pend[i]);
         st[i].advance();
       }
@@ -131,13 +157,24 @@ using namespace tdsa;
 
 extern "C" {
 
+// Tile height: 256 when N allows (halves staged bytes per FLOP), else 128.
+static int tn_nw(int N) {
+  int nw = (N % 256 == 0) ? 8 : 4;
+  if (const char* v = getenv("TDSA_GEMM_TN_NW")) {
+    int e = atoi(v);
+    if ((e == 4 || e == 8) && N % (e * 32) == 0) nw = e;
+  }
+  return nw;
+}
+
 // Returns the split count the launcher will use (for the caller to decide
 // whether the fp32 buffer needs zeroing); <=0 means shape unsupported.
 int tdsa_gemm_tn_splits(long long M, int N, int K) {
   if (M <= 0 || (M & 63) || (N & 127) || (K & 127)) return 0;
-  const long long tiles = (long long)(N >> 7) * (K >> 7);
-  // enough workgroups to fill 256 CUs several times over
-  long long want = (2048 + tiles - 1) / tiles;
+  const int nw = tn_nw(N);
+  const long long tiles = (long long)(N / (nw * 32)) * (K >> 7);
+  // enough workgroups to give every CU a few, without excess atomics
+  long long want = (1024 + tiles - 1) / tiles;
   long long maxs = M >> 6;  // at least one 64-chunk per split
   if (want > maxs) want = maxs;
   if (const char* v = getenv("TDSA_GEMM_TN_SPLITS")) want = atoll(v);
@@ -149,17 +186,22 @@ hipError_t tdsa_gemm_tn(const void* dy, const void* x, float* dw, long long M,
                         int N, int K, hipStream_t stream) {
   const int splits = tdsa_gemm_tn_splits(M, N, K);
   if (splits <= 0) return hipErrorInvalidValue;
+  const int nw = tn_nw(N);
   const long long chunks_per_split = ((M >> 6) + splits - 1) / splits;
   const long long m_per_split = chunks_per_split << 6;
-  dim3 grid((N >> 7) * (K >> 7), splits);
-  if (splits == 1)
-    hipLaunchKernelGGL((gemm_tn_kernel<false>), grid, dim3(256), 0, stream,
-                       (const bf16*)dy, (const bf16*)x, dw, M, N, K,
-                       m_per_split);
-  else
-    hipLaunchKernelGGL((gemm_tn_kernel<true>), grid, dim3(256), 0, stream,
-                       (const bf16*)dy, (const bf16*)x, dw, M, N, K,
-                       m_per_split);
+  int remap = 1;
+  if (const char* v = getenv("TDSA_GEMM_TN_REMAP")) remap = atoi(v);
+  dim3 grid((N / (nw * 32)) * (K >> 7), splits);
+#define TN_LAUNCH(NWV, AT)                                                  \
+  hipLaunchKernelGGL((gemm_tn_kernel<NWV, AT>), grid, dim3(NWV * WAVE), 0,  \
+                     stream, (const bf16*)dy, (const bf16*)x, dw, M, N, K,  \
+                     m_per_split, remap)
+  if (nw == 8) {
+    if (splits == 1) TN_LAUNCH(8, false); else TN_LAUNCH(8, true);
+  } else {
+    if (splits == 1) TN_LAUNCH(4, false); else TN_LAUNCH(4, true);
+  }
+#undef TN_LAUNCH
   return hipGetLastError();
 }
 

@@ -55,7 +55,14 @@ struct Stage {
   int dst[REPS];
   long long step;  // elements to advance per tile
 
+  DEV_INLINE Stage() {}
+
   DEV_INLINE Stage(const bf16* g, int tid, long long t_stride, int rows = 64) {
+    init(g, tid, t_stride, rows);
+  }
+
+  DEV_INLINE void init(const bf16* g, int tid, long long t_stride,
+                       int rows = 64) {
     step = (long long)rows * t_stride;
 #pragma unroll
     for (int rep = 0; rep < REPS; ++rep) {
