@@ -157,9 +157,11 @@ using namespace tdsa;
 
 extern "C" {
 
-// Tile height: 256 when N allows (halves staged bytes per FLOP), else 128.
+// Tile height: 4 waves / 128 rows measured BEST (the 8-wave 256-row tile
+// variant lost 10-20% on every shape despite doubled staging intensity —
+// gpurun gemm_tn_v2 vs v1 sweeps); 8 kept behind the env for re-testing.
 static int tn_nw(int N) {
-  int nw = (N % 256 == 0) ? 8 : 4;
+  int nw = 4;
   if (const char* v = getenv("TDSA_GEMM_TN_NW")) {
     int e = atoi(v);
     if ((e == 4 || e == 8) && N % (e * 32) == 0) nw = e;
@@ -173,8 +175,9 @@ int tdsa_gemm_tn_splits(long long M, int N, int K) {
   if (M <= 0 || (M & 63) || (N & 127) || (K & 127)) return 0;
   const int nw = tn_nw(N);
   const long long tiles = (long long)(N / (nw * 32)) * (K >> 7);
-  // enough workgroups to give every CU a few, without excess atomics
-  long long want = (1024 + tiles - 1) / tiles;
+  // enough workgroups to fill 256 CUs several times over (2048 target
+  // measured better than 1024 on the small-output shapes)
+  long long want = (2048 + tiles - 1) / tiles;
   long long maxs = M >> 6;  // at least one 64-chunk per split
   if (want > maxs) want = maxs;
   if (const char* v = getenv("TDSA_GEMM_TN_SPLITS")) want = atoll(v);
@@ -189,7 +192,7 @@ hipError_t tdsa_gemm_tn(const void* dy, const void* x, float* dw, long long M,
   const int nw = tn_nw(N);
   const long long chunks_per_split = ((M >> 6) + splits - 1) / splits;
   const long long m_per_split = chunks_per_split << 6;
-  int remap = 1;
+  int remap = 0;  // measured: no effect on any shape (see profiles/)
   if (const char* v = getenv("TDSA_GEMM_TN_REMAP")) remap = atoi(v);
   dim3 grid((N / (nw * 32)) * (K >> 7), splits);
 #define TN_LAUNCH(NWV, AT)                                                  \

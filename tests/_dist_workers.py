@@ -182,6 +182,62 @@ def grad_accumulation(rank, world):
     return lin.weight.grad.clone()
 
 
+def fused_lmhead_strategy_losses(rank, world, strategy):
+    """Strategy training with fused_lm_head=True: dW publication must go
+    through publish_weight_grad into the strategy's collective."""
+    import tiny_deepspeed_amd as tdsa
+    from tiny_deepspeed_amd.models import GPTConfig, GPT2Model
+
+    cfg = GPTConfig(fused_lm_head=True, **CFG)
+    torch.manual_seed(0)
+    model = GPT2Model(cfg)
+    x, y = batch()
+    if strategy == "ddp":
+        model = tdsa.DDP(model)
+        opt = tdsa.DDPAdamW(model.named_parameters(), lr=1e-3,
+                            weight_decay=0.01)
+    elif strategy == "zero2flat":
+        model = tdsa.Zero2Flat(model)
+        opt = tdsa.Zero2FlatAdamW(model, lr=1e-3, weight_decay=0.01)
+    else:
+        with torch.device("meta"):
+            meta = GPT2Model(cfg)
+        parts, _ = tdsa.partition_tensors(
+            OrderedDict(meta.named_parameters()), ["cpu"] * world)
+        wrap = {"zero2": tdsa.Zero2, "zero3": tdsa.Zero3}[strategy]
+        optc = {"zero2": tdsa.Zero2AdamW, "zero3": tdsa.Zero3AdamW}[strategy]
+        model = wrap(model, parts)
+        opt = optc(model.named_parameters(), lr=1e-3, weight_decay=0.01,
+                   param_part_table=parts, ranks_map=["cpu"] * world)
+    losses = []
+    for _ in range(ITERS):
+        model.require_backward_grad_sync = True
+        logits, loss = model(x, y)
+        assert logits is None  # fused path ran
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    return losses
+
+
+def fused_lmhead_single_losses():
+    import tiny_deepspeed_amd as tdsa
+    from tiny_deepspeed_amd.models import GPTConfig, GPT2Model
+
+    cfg = GPTConfig(fused_lm_head=True, **CFG)
+    torch.manual_seed(0)
+    model = tdsa.Single(GPT2Model(cfg))
+    opt = tdsa.AdamW(model.named_parameters(), lr=1e-3, weight_decay=0.01)
+    x, y = batch()
+    losses = []
+    for _ in range(ITERS):
+        _, loss = model(x, y)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    return losses
+
+
 def zero2flat_losses(rank, world, bucket_bytes=32 << 20):
     """Flat-bucket ZeRO-2: returns (losses, collective counts, n_buckets,
     n_params) so the test can assert loss parity AND the O(#params) ->

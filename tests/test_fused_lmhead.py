@@ -63,7 +63,7 @@ def test_fused_lmhead_bias_rejected():
 def test_model_uses_fused_path_and_trains(monkeypatch):
     monkeypatch.setenv("TDSA_CE_CHUNK", "16")
     cfg = GPTConfig(n_layer=2, n_head=2, n_embd=32, block_size=32,
-                    vocab_size=64)
+                    vocab_size=64, fused_lm_head=True)
     torch.manual_seed(0)
     model = Single(GPT2Model(cfg))
     opt = AdamW(model.named_parameters(), lr=1e-3)
@@ -82,6 +82,20 @@ def test_model_uses_fused_path_and_trains(monkeypatch):
     logits, loss = model(x)
     assert logits is not None and loss is None
     assert logits.shape == (2, 32, 64)
+
+
+@pytest.mark.parametrize("strategy", ["ddp", "zero2", "zero3", "zero2flat"])
+def test_fused_lmhead_strategy_parity_world2(strategy):
+    """dW of the fused path must flow through each strategy's collective
+    (publish_weight_grad): world-2 losses == single-device fused losses."""
+    from tests.dist_utils import run_distributed
+    from tests import _dist_workers as w
+
+    single = w.fused_lmhead_single_losses()
+    results = run_distributed(w.fused_lmhead_strategy_losses, world=2,
+                              args=(strategy,))
+    for rank, losses in results.items():
+        assert losses == pytest.approx(single, rel=1e-4), (rank, losses)
 
 
 def test_fused_path_matches_unfused_model(monkeypatch):

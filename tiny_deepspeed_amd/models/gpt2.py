@@ -34,9 +34,13 @@ class GPTConfig:
     # "fused": CDNA4 flash-style kernel (composite torch on CPU);
     # "math": composite torch attention everywhere.
     attention: str = "fused"
-    # row-chunked fused lm_head projection + cross-entropy (loss-only path
-    # returns logits=None; inference with targets=None is unaffected)
-    fused_lm_head: bool = True
+    # row-chunked fused lm_head projection + cross-entropy: the full
+    # (B*T, V) logits tensor is never materialized (-2.75 GB peak at
+    # gpt2-medium b32) at the cost of recomputing chunk logits in backward
+    # (+4 ms/step, ~3.6% — same-box A/B in BASELINE.md). Default off: the
+    # throughput headline wins; flip on when memory-bound. Loss-only path
+    # returns logits=None; inference with targets=None is unaffected.
+    fused_lm_head: bool = False
 
     @classmethod
     def gpt2_small(cls, **kw):
