@@ -40,6 +40,7 @@
 #include "mfma.h"
 
 #include <cstdlib>
+#include <type_traits>
 
 namespace tdsa {
 
@@ -394,24 +395,33 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
         dp_acc = MFMA32(lds_read16(lds_do, AF_OFF(t2, s)), v_frag[s], dp_acc);
       }
       __builtin_amdgcn_s_setprio(0);
+      // MASK as a compile-time split: `diag` is wave-uniform but inside the
+      // unrolled loop the single-version ternary was if-converted into
+      // cmp+cndmask+add on EVERY tile (~96 VALU/tile measured in the .s);
+      // only ~1 of 16 tiles is diagonal.
+      auto consume = [&](auto MASK) {
 #pragma unroll
-      for (int r1 = 0; r1 < 4; ++r1) {
-        float p[4], dsv[4];
+        for (int r1 = 0; r1 < 4; ++r1) {
+          float p[4], dsv[4];
 #pragma unroll
-        for (int e = 0; e < 4; ++e) {
-          const int r = 4 * r1 + e;
-          const int lrow = t2 * 32 + crow(r, h32);
-          const int qrow = q0 + lrow;
-          float pp = fast_exp2(s_acc[r] - lds_lse[lrow]);
-          if (diag) pp = (key_me <= qrow) ? pp : 0.f;
-          p[e] = pp;
-          dsv[e] = pp * (dp_acc[r] - lds_dlt[lrow]);  // scale in epilogue
+          for (int e = 0; e < 4; ++e) {
+            const int r = 4 * r1 + e;
+            const int lrow = t2 * 32 + crow(r, h32);
+            float pp = fast_exp2(s_acc[r] - lds_lse[lrow]);
+            if (MASK.value) pp = (key_me <= q0 + lrow) ? pp : 0.f;
+            p[e] = pp;
+            dsv[e] = pp * (dp_acc[r] - lds_dlt[lrow]);  // scale in epilogue
+          }
+          P.wA[t2][r1] = pack2(p[0], p[1]);
+          P.wB[t2][r1] = pack2(p[2], p[3]);
+          dS.wA[t2][r1] = pack2(dsv[0], dsv[1]);
+          dS.wB[t2][r1] = pack2(dsv[2], dsv[3]);
         }
-        P.wA[t2][r1] = pack2(p[0], p[1]);
-        P.wB[t2][r1] = pack2(p[2], p[3]);
-        dS.wA[t2][r1] = pack2(dsv[0], dsv[1]);
-        dS.wB[t2][r1] = pack2(dsv[2], dsv[3]);
-      }
+      };
+      if (diag)
+        consume(std::true_type{});
+      else
+        consume(std::false_type{});
     }
 
     // dV[key][d] += P^T dO ; dK[key][d] += dS^T Q  (A-frag k = qrow)
@@ -564,13 +574,22 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
       }
       __builtin_amdgcn_s_setprio(0);
       float dsv[16];
+      // wave-uniform mask split (same rationale as the dkv kernel)
+      auto consume = [&](auto MASK) {
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int key = key0 + t2 * 32 + crow(r, h32);
-        float pp = fast_exp2(s_acc[r] - lse2_me);
-        if (diag) pp = (key <= row_me) ? pp : 0.f;
-        dsv[r] = pp * (dp_acc[r] - dlt_me);
-      }
+        for (int r = 0; r < 16; ++r) {
+          float pp = fast_exp2(s_acc[r] - lse2_me);
+          if (MASK.value) {
+            const int key = key0 + t2 * 32 + crow(r, h32);
+            pp = (key <= row_me) ? pp : 0.f;
+          }
+          dsv[r] = pp * (dp_acc[r] - dlt_me);
+        }
+      };
+      if (diag)
+        consume(std::true_type{});
+      else
+        consume(std::false_type{});
 #pragma unroll
       for (int r1 = 0; r1 < 4; ++r1) {
         dS.wA[t2][r1] = pack2(dsv[4 * r1], dsv[4 * r1 + 1]);
