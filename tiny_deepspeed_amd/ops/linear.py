@@ -19,6 +19,8 @@ hand-written split-M MFMA kernel as a tuner candidate so the question is
 settled by measurement per shape, not by assertion.
 """
 
+import os
+
 import torch
 
 from . import _ext
@@ -53,6 +55,8 @@ def dw_hip(dy2, x2):
 
 
 def _dw_hip_supported(dy2, x2):
+    if os.environ.get("TDSA_GEMM_TN", "1") == "0":
+        return False
     if not _ext.ext_available() or not hasattr(_ext.get_ext(), "gemm_tn"):
         return False
     # kernel contract (csrc/kernels/gemm_tn.hip): bf16, M % 64 == 0,
