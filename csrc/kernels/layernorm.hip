@@ -101,12 +101,18 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res
   }
 }
 
-// Backward dx + per-block dw/db stripe partials. Grid = G blocks; block g
-// handles rows g, g+G, ... and accumulates its dw/db into pdw[g*N..], so the
-// stripe write needs no atomics at all.
+// Backward dx (+ optionally the dw/db stripe partials). Grid = G blocks;
+// block g handles rows g, g+G, ...; with STRIPES it accumulates dw/db into
+// pdw[g*N..] (no atomics).
+// Measured split (r2): the fused form chains rows with two block-wide
+// reduction barriers each, capping dx at ~2.1 TB/s while the fwd kernel
+// reaches 3.1. The launcher now runs dx at FULL grid (one row per block,
+// STRIPES=false) and a separate barrier-free streaming kernel
+// (ln_dwdb_accum_kernel) produces the stripes — one extra dy/x read buys
+// unchained dx rows.
 // HAS_DH: dx += dh (gradient of the residual-stream output h), fusing the
 // backward-side elementwise add of the residual connection.
-template <typename T, int MAXITER, bool HAS_DH>
+template <typename T, int MAXITER, bool HAS_DH, bool STRIPES = true>
 __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__ dh,
                                  const T* __restrict__ x,
                                  const T* __restrict__ w, const float* __restrict__ mean,
@@ -147,8 +153,10 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__
           float wdy = VT::get(wv, k) * d;
           c1 += xh * wdy;
           c2 += wdy;
-          accdw[it][k] += d * xh;
-          accdb[it][k] += d;
+          if (STRIPES) {
+            accdw[it][k] += d * xh;
+            accdb[it][k] += d;
+          }
         }
       }
     }
@@ -180,9 +188,9 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__
     }
   }
   // stripe write: block-owned rows of pdw/pdb
-  float* sdw = pdw + (long long)blockIdx.x * N;
-  float* sdb = pdb + (long long)blockIdx.x * N;
-  {
+  if (STRIPES) {
+    float* sdw = pdw + (long long)blockIdx.x * N;
+    float* sdb = pdb + (long long)blockIdx.x * N;
     int it = 0;
     for (int i = tid * W; i < N; i += nth * W, ++it) {
 #pragma unroll
@@ -190,6 +198,56 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__
         sdw[i + k] = accdw[it][k];
         sdb[i + k] = accdb[it][k];
       }
+    }
+  }
+}
+
+// Streaming dw/db stripe accumulation (no barriers: the per-column sums
+// need no cross-thread reduction). Block g chains rows g, g+G, ... exactly
+// like the fused form, but with nothing serializing the row loop it runs
+// at the HBM read floor.
+template <typename T, int MAXITER>
+__global__ void ln_dwdb_accum_kernel(const T* __restrict__ dy,
+                                     const T* __restrict__ x,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ rstd,
+                                     float* __restrict__ pdw,
+                                     float* __restrict__ pdb, int M, int N) {
+  using VT = VecTraits<T>;
+  constexpr int W = VT::W;
+  const int tid = threadIdx.x;
+  const int nth = blockDim.x;
+  float accdw[MAXITER][W];
+  float accdb[MAXITER][W];
+#pragma unroll
+  for (int it = 0; it < MAXITER; ++it)
+#pragma unroll
+    for (int k = 0; k < W; ++k) accdw[it][k] = accdb[it][k] = 0.f;
+  for (int row = blockIdx.x; row < M; row += gridDim.x) {
+    const T* dyr = dy + (long long)row * N;
+    const T* xr = x + (long long)row * N;
+    const float mu = mean[row];
+    const float rs = rstd[row];
+    int it = 0;
+    for (int i = tid * W; i < N; i += nth * W, ++it) {
+      typename VT::V dv = VT::load(dyr + i);
+      typename VT::V xv = VT::load(xr + i);
+#pragma unroll
+      for (int k = 0; k < W; ++k) {
+        float d = VT::get(dv, k);
+        accdw[it][k] += d * (VT::get(xv, k) - mu) * rs;
+        accdb[it][k] += d;
+      }
+    }
+  }
+  float* sdw = pdw + (long long)blockIdx.x * N;
+  float* sdb = pdb + (long long)blockIdx.x * N;
+  int it = 0;
+  for (int i = tid * W; i < N; i += nth * W, ++it) {
+#pragma unroll
+    for (int k = 0; k < W; ++k) {
+      sdw[i + k] = accdw[it][k];
+      sdb[i + k] = accdb[it][k];
     }
   }
 }
@@ -423,16 +481,36 @@ int tdsa_ln_bwd_dx_stripes(int M) {
 }
 
 // dh: optional residual-stream gradient added into dx (nullptr for plain).
+// Default is the SPLIT scheme (dx at full grid + streaming stripe accum;
+// see ln_bwd_dx_kernel comment); TDSA_LN_SPLIT=0 restores the fused form.
 hipError_t tdsa_ln_bwd_dx(const void* dy, const void* dh, const void* x,
                           const void* w, const float* mean, const float* rstd,
                           void* dx, float* pdw, float* pdb, int M, int N,
                           int is_bf16, hipStream_t stream) {
   const int block = ln_block(N, is_bf16 ? 8 : 4);
   const int grid = tdsa_ln_bwd_dx_stripes(M);
-#define LN_BWD(T, HASD)                                                       \
-  hipLaunchKernelGGL((ln_bwd_dx_kernel<T, 2, HASD>), dim3(grid), dim3(block), \
-                     0, stream, (const T*)dy, (const T*)dh, (const T*)x,      \
-                     (const T*)w, mean, rstd, (T*)dx, pdw, pdb, M, N)
+  int split = 1;
+  if (const char* v = getenv("TDSA_LN_SPLIT")) split = atoi(v);
+  int dx_cap = 32768;
+  if (const char* v = getenv("TDSA_LN_GRID")) dx_cap = atoi(v);
+  const int dx_grid = split ? ((M < dx_cap) ? M : dx_cap) : grid;
+#define LN_BWD(T, HASD)                                                        \
+  do {                                                                         \
+    if (split) {                                                               \
+      hipLaunchKernelGGL((ln_bwd_dx_kernel<T, 2, HASD, false>), dim3(dx_grid), \
+                         dim3(block), 0, stream, (const T*)dy, (const T*)dh,   \
+                         (const T*)x, (const T*)w, mean, rstd, (T*)dx, pdw,    \
+                         pdb, M, N);                                           \
+      hipLaunchKernelGGL((ln_dwdb_accum_kernel<T, 2>), dim3(grid),             \
+                         dim3(block), 0, stream, (const T*)dy, (const T*)x,    \
+                         mean, rstd, pdw, pdb, M, N);                          \
+    } else {                                                                   \
+      hipLaunchKernelGGL((ln_bwd_dx_kernel<T, 2, HASD, true>), dim3(grid),     \
+                         dim3(block), 0, stream, (const T*)dy, (const T*)dh,   \
+                         (const T*)x, (const T*)w, mean, rstd, (T*)dx, pdw,    \
+                         pdb, M, N);                                           \
+    }                                                                          \
+  } while (0)
 #define LN_BWD_WIDE(T, HASD)                                                  \
   hipLaunchKernelGGL((ln_bwd_dx_wide_kernel<T, HASD>), dim3(grid),            \
                      dim3(1024), 0, stream, (const T*)dy, (const T*)dh,       \
