@@ -202,25 +202,91 @@ __global__ void adamw_multi_kernel(const AdamTensorDesc* __restrict__ descs,
   }
 }
 
+DEV_INLINE float sgd_math(float gv, float& pv, float* buf, long long i,
+                          float lr, float momentum, float dampening, float wd,
+                          int nesterov, int maximize, int first_step) {
+  if (maximize) gv = -gv;
+  if (wd != 0.0f) gv += wd * pv;
+  if (buf) {
+    float b = first_step ? gv : buf[i] * momentum + (1.0f - dampening) * gv;
+    buf[i] = b;
+    gv = nesterov ? (gv + momentum * b) : b;
+  }
+  pv -= lr * gv;
+  return pv;
+}
+
+// 4 elements/lane with explicit dwordx4 state accesses (same rationale as
+// adamw_span_vec; the scalar form compiled to flat_load_dword streams).
+template <typename PT, typename GT, bool HASB, bool HASM>
+DEV_INLINE void sgd_span_vec(PT* p, const GT* g, float* buf, float* master,
+                             long long start, long long end, float lr,
+                             float momentum, float dampening, float wd,
+                             int nesterov, int maximize, int first_step) {
+  const long long n4 = (end - start) / 4;
+  for (long long q = threadIdx.x; q < n4; q += blockDim.x) {
+    const long long i = start + q * 4;
+    typename Vec4<GT>::V gv = Vec4<GT>::load(g + i);
+    f32x4 pv;
+    if (HASM) {
+      pv = *reinterpret_cast<const f32x4*>(master + i);
+    } else {
+      typename Vec4<PT>::V praw = Vec4<PT>::load(p + i);
+#pragma unroll
+      for (int k = 0; k < 4; ++k) pv[k] = Vec4<PT>::get(praw, k);
+    }
+    f32x4 bv;
+    if (HASB) bv = *reinterpret_cast<const f32x4*>(buf + i);
+    typename Vec4<PT>::V pout;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      float gk = Vec4<GT>::get(gv, k);
+      if (maximize) gk = -gk;
+      float pk = pv[k];
+      if (wd != 0.0f) gk += wd * pk;
+      if (HASB) {
+        float b = first_step ? gk : bv[k] * momentum + (1.0f - dampening) * gk;
+        bv[k] = b;
+        gk = nesterov ? (gk + momentum * b) : b;
+      }
+      pk -= lr * gk;
+      pv[k] = pk;
+      Vec4<PT>::set(pout, k, pk);
+    }
+    if (HASB) *reinterpret_cast<f32x4*>(buf + i) = bv;
+    if (HASM) *reinterpret_cast<f32x4*>(master + i) = pv;
+    Vec4<PT>::store(p + i, pout);
+  }
+  for (long long i = start + n4 * 4 + threadIdx.x; i < end; i += blockDim.x) {
+    float pv = HASM ? master[i] : (float)p[i];
+    pv = sgd_math((float)g[i], pv, HASB ? buf : nullptr, i, lr, momentum,
+                  dampening, wd, nesterov, maximize, first_step);
+    if (HASM) master[i] = pv;
+    p[i] = (PT)pv;
+  }
+}
+
 template <typename PT, typename GT>
 DEV_INLINE void sgd_update_span(PT* p, const GT* g, float* buf, float* master,
                                 long long start, long long end, float lr,
                                 float momentum, float dampening, float wd,
                                 int nesterov, int maximize, int first_step) {
-  for (long long i = start + threadIdx.x; i < end; i += blockDim.x) {
-    float gv = (float)g[i];
-    if (maximize) gv = -gv;
-    float pv = master ? master[i] : (float)p[i];
-    if (wd != 0.0f) gv += wd * pv;
-    if (buf) {
-      float b = first_step ? gv : buf[i] * momentum + (1.0f - dampening) * gv;
-      buf[i] = b;
-      gv = nesterov ? (gv + momentum * b) : b;
-    }
-    pv -= lr * gv;
-    if (master) master[i] = pv;
-    p[i] = (PT)pv;
-  }
+  if (buf && master)
+    sgd_span_vec<PT, GT, true, true>(p, g, buf, master, start, end, lr,
+                                     momentum, dampening, wd, nesterov,
+                                     maximize, first_step);
+  else if (buf)
+    sgd_span_vec<PT, GT, true, false>(p, g, buf, master, start, end, lr,
+                                      momentum, dampening, wd, nesterov,
+                                      maximize, first_step);
+  else if (master)
+    sgd_span_vec<PT, GT, false, true>(p, g, buf, master, start, end, lr,
+                                      momentum, dampening, wd, nesterov,
+                                      maximize, first_step);
+  else
+    sgd_span_vec<PT, GT, false, false>(p, g, buf, master, start, end, lr,
+                                       momentum, dampening, wd, nesterov,
+                                       maximize, first_step);
 }
 
 struct SgdTensorDesc {
