@@ -89,6 +89,27 @@ def test_final_tune_freezes_unseen_keys_to_default(fake_ext):
     assert tuner.choices() == {}  # no tuning after freeze; default ran
 
 
+def test_tuner_cache_roundtrip(fake_ext, tmp_path):
+    """save_cache/load_cache: a second tuner adopts the first one's winners
+    without re-measuring (used for tuning-noise-free rocprof captures)."""
+    path = str(tmp_path / "tuner.json")
+    t1 = RuntimeAutoTuner(warmup=1, iters=3)
+    dy = torch.randn(64, 32)
+    linear_ops.linear_bias_grad(dy, tuner=t1)
+    assert list(t1.choices().values()) == ["db_torch"]
+    t1.save_cache(path)
+
+    t2 = RuntimeAutoTuner(warmup=1, iters=3)
+    t2.load_cache(path)
+    calls = []
+    orig_time = t2._time_one
+    t2._time_one = lambda *a, **k: (calls.append(1), orig_time(*a, **k))[1]
+    out = linear_ops.linear_bias_grad(dy, tuner=t2)
+    torch.testing.assert_close(out, dy.sum(dim=0))
+    assert calls == []  # no measurement ran
+    assert list(t2.choices().values()) == ["db_torch"]
+
+
 def test_default_tuner_disabled_by_env(monkeypatch):
     import tiny_deepspeed_amd.ops.autotuner as at
     monkeypatch.setenv("TDSA_AUTOTUNE", "0")
