@@ -34,7 +34,10 @@ template <> struct VecTraits<bf16> {
 // HAS_RES: fused residual add — h = x + res is computed in-kernel, written
 // out (the residual stream) and normalized, saving the separate elementwise
 // add pass + launch per LayerNorm site (2 per transformer block).
-template <typename T, bool HAS_RES>
+// ONE_WAVE: 64-thread blocks, one wave per row — the two per-row reductions
+// become pure wave shuffles with NO __syncthreads / LDS at all (needs a
+// deeper register cache: MAXITER up to 4 covers N<=2048 bf16 / 1024 fp32).
+template <typename T, bool HAS_RES, int MAXITER = 2, bool ONE_WAVE = false>
 __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res,
                               T* __restrict__ h, const T* __restrict__ w,
                               const T* __restrict__ b, T* __restrict__ y,
@@ -42,7 +45,6 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res
                               int M, int N, float eps) {
   using VT = VecTraits<T>;
   constexpr int W = VT::W;
-  constexpr int MAXITER = 2;  // rows cached in registers between passes
   __shared__ float scratch[2 * 1024 / WAVE];
   const int tid = threadIdx.x;
   const int nth = blockDim.x;
@@ -74,7 +76,12 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res
         }
       }
     }
-    block_sum2(s, ss, scratch);
+    if (ONE_WAVE) {
+      s = wave_sum(s);
+      ss = wave_sum(ss);
+    } else {
+      block_sum2(s, ss, scratch);
+    }
     const float mu = s / N;
     const float var = fmaxf(ss / N - mu * mu, 0.0f);
     const float rs = rsqrtf(var + eps);
@@ -112,7 +119,8 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res
 // unchained dx rows.
 // HAS_DH: dx += dh (gradient of the residual-stream output h), fusing the
 // backward-side elementwise add of the residual connection.
-template <typename T, int MAXITER, bool HAS_DH, bool STRIPES = true>
+template <typename T, int MAXITER, bool HAS_DH, bool STRIPES = true,
+          bool ONE_WAVE = false>
 __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__ dh,
                                  const T* __restrict__ x,
                                  const T* __restrict__ w, const float* __restrict__ mean,
@@ -160,7 +168,12 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__
         }
       }
     }
-    block_sum2(c1, c2, scratch);
+    if (ONE_WAVE) {
+      c1 = wave_sum(c1);
+      c2 = wave_sum(c2);
+    } else {
+      block_sum2(c1, c2, scratch);
+    }
     c1 /= N;
     c2 /= N;
     T* dxr = dx + (long long)row * N;
@@ -442,30 +455,43 @@ hipError_t tdsa_ln_fwd(const void* x, const void* res, void* h, const void* w,
   int cap = 32768;
   if (const char* v = getenv("TDSA_LN_GRID")) cap = atoi(v);
   const int grid = (M < cap) ? M : cap;
+  // one-wave rows (no barriers) when the row fits a 64-lane register cache
+  int wave_ok = 1;
+  if (const char* v = getenv("TDSA_LN_WAVE")) wave_ok = atoi(v);
 #define LN_FWD(T, HASR)                                                       \
   hipLaunchKernelGGL((ln_fwd_kernel<T, HASR>), dim3(grid), dim3(block), 0,    \
                      stream, (const T*)x, (const T*)res, (T*)h, (const T*)w,  \
                      (const T*)b, (T*)y, mean, rstd, M, N, eps)
+#define LN_FWD_OW(T, HASR)                                                    \
+  hipLaunchKernelGGL((ln_fwd_kernel<T, HASR, 4, true>), dim3(grid),           \
+                     dim3(WAVE), 0, stream, (const T*)x, (const T*)res,       \
+                     (T*)h, (const T*)w, (const T*)b, (T*)y, mean, rstd, M,   \
+                     N, eps)
 #define LN_FWD_WIDE(T, HASR)                                                  \
   hipLaunchKernelGGL((ln_fwd_wide_kernel<T, HASR>), dim3(grid), dim3(1024),   \
                      0, stream, (const T*)x, (const T*)res, (T*)h,            \
                      (const T*)w, (const T*)b, (T*)y, mean, rstd, M, N, eps)
   if (is_bf16) {
     if (N % 8) return hipErrorInvalidValue;  // 16B row-base alignment
-    if (N > 2 * block * 8) {
+    if (wave_ok && N <= 4 * WAVE * 8) {
+      if (res) LN_FWD_OW(bf16, true); else LN_FWD_OW(bf16, false);
+    } else if (N > 2 * block * 8) {
       if (res) LN_FWD_WIDE(bf16, true); else LN_FWD_WIDE(bf16, false);
     } else {
       if (res) LN_FWD(bf16, true); else LN_FWD(bf16, false);
     }
   } else {
     if (N % 4) return hipErrorInvalidValue;
-    if (N > 2 * block * 4) {
+    if (wave_ok && N <= 4 * WAVE * 4) {
+      if (res) LN_FWD_OW(float, true); else LN_FWD_OW(float, false);
+    } else if (N > 2 * block * 4) {
       if (res) LN_FWD_WIDE(float, true); else LN_FWD_WIDE(float, false);
     } else {
       if (res) LN_FWD(float, true); else LN_FWD(float, false);
     }
   }
 #undef LN_FWD
+#undef LN_FWD_OW
 #undef LN_FWD_WIDE
   return hipGetLastError();
 }
@@ -494,9 +520,20 @@ hipError_t tdsa_ln_bwd_dx(const void* dy, const void* dh, const void* x,
   int dx_cap = 32768;
   if (const char* v = getenv("TDSA_LN_GRID")) dx_cap = atoi(v);
   const int dx_grid = split ? ((M < dx_cap) ? M : dx_cap) : grid;
+  int wave_ok = 1;
+  if (const char* v = getenv("TDSA_LN_WAVE")) wave_ok = atoi(v);
+  const int ow = wave_ok && N <= 4 * WAVE * (is_bf16 ? 8 : 4);
 #define LN_BWD(T, HASD)                                                        \
   do {                                                                         \
-    if (split) {                                                               \
+    if (split && ow) {                                                         \
+      hipLaunchKernelGGL((ln_bwd_dx_kernel<T, 4, HASD, false, true>),          \
+                         dim3(dx_grid), dim3(WAVE), 0, stream, (const T*)dy,   \
+                         (const T*)dh, (const T*)x, (const T*)w, mean, rstd,   \
+                         (T*)dx, pdw, pdb, M, N);                              \
+      hipLaunchKernelGGL((ln_dwdb_accum_kernel<T, 2>), dim3(grid),             \
+                         dim3(block), 0, stream, (const T*)dy, (const T*)x,    \
+                         mean, rstd, pdw, pdb, M, N);                          \
+    } else if (split) {                                                        \
       hipLaunchKernelGGL((ln_bwd_dx_kernel<T, 2, HASD, false>), dim3(dx_grid), \
                          dim3(block), 0, stream, (const T*)dy, (const T*)dh,   \
                          (const T*)x, (const T*)w, mean, rstd, (T*)dx, pdw,    \
