@@ -455,8 +455,10 @@ hipError_t tdsa_ln_fwd(const void* x, const void* res, void* h, const void* w,
   int cap = 32768;
   if (const char* v = getenv("TDSA_LN_GRID")) cap = atoi(v);
   const int grid = (M < cap) ? M : cap;
-  // one-wave rows (no barriers) when the row fits a 64-lane register cache
-  int wave_ok = 1;
+  // one-wave rows (no barriers): measured 2-3x SLOWER than the 2-wave
+  // block form (64-lane blocks lose memory-level parallelism; the barrier
+  // was never the bottleneck) — kept behind TDSA_LN_WAVE=1 as a record
+  int wave_ok = 0;
   if (const char* v = getenv("TDSA_LN_WAVE")) wave_ok = atoi(v);
 #define LN_FWD(T, HASR)                                                       \
   hipLaunchKernelGGL((ln_fwd_kernel<T, HASR>), dim3(grid), dim3(block), 0,    \
@@ -520,7 +522,7 @@ hipError_t tdsa_ln_bwd_dx(const void* dy, const void* dh, const void* x,
   int dx_cap = 32768;
   if (const char* v = getenv("TDSA_LN_GRID")) dx_cap = atoi(v);
   const int dx_grid = split ? ((M < dx_cap) ? M : dx_cap) : grid;
-  int wave_ok = 1;
+  int wave_ok = 0;  // measured worse (see tdsa_ln_fwd comment)
   if (const char* v = getenv("TDSA_LN_WAVE")) wave_ok = atoi(v);
   const int ow = wave_ok && N <= 4 * WAVE * (is_bf16 ? 8 : 4);
 #define LN_BWD(T, HASD)                                                        \
