@@ -164,6 +164,9 @@ def zero3_meta_init(rank, world):
         loss.backward()
         opt.step()
         losses.append(loss.item())
+    # meta materialization must honor the model's init scheme (std 0.02):
+    # the nn-default inits started near loss ~10 instead of ~ln(vocab)
+    assert losses[0] < 6.0, losses
     return losses
 
 
