@@ -64,6 +64,27 @@ def test_rccl_bucketed_broadcast():
         torch.testing.assert_close(t, r)
 
 
+def test_rccl_flat_collectives():
+    """reduce_scatter_avg / all_gather_flat (flat ZeRO-2) through real RCCL
+    enqueue at forced world-1: shard == whole, values round-trip."""
+    comm = CommContext()
+    flat = torch.randn(1 << 20, device="cuda")
+    ref = flat.clone()
+    shard = torch.empty_like(flat)
+    comm.reduce_scatter_avg(shard, flat)
+    comm.sync()
+    torch.testing.assert_close(shard, ref)
+    out = torch.empty_like(flat)
+    comm.all_gather_flat(out, shard)
+    comm.sync()
+    torch.testing.assert_close(out, ref)
+    # in-place aliasing form (the engine's layout: shard is a view of out)
+    out2 = ref.clone()
+    comm.all_gather_flat(out2, out2[: out2.numel()])
+    comm.sync()
+    torch.testing.assert_close(out2, ref)
+
+
 def test_rccl_grad_release_lifetime():
     """ZeRO-2-style release under RCCL: enqueue reduce on the comm stream,
     drop the last host reference, immediately allocate/compute over the
