@@ -175,9 +175,10 @@ int tdsa_gemm_tn_splits(long long M, int N, int K) {
   if (M <= 0 || (M & 63) || (N & 127) || (K & 127)) return 0;
   const int nw = tn_nw(N);
   const long long tiles = (long long)(N / (nw * 32)) * (K >> 7);
-  // enough workgroups to fill 256 CUs several times over (2048 target
-  // measured better than 1024 on the small-output shapes)
-  long long want = (2048 + tiles - 1) / tiles;
+  // ~512 total workgroups (2 per CU): the splits sweep measured 8 splits
+  // at 64 tiles (512 wgs) 25% faster than 32 splits — more splits just
+  // multiply the fp32 atomic traffic once every CU has work
+  long long want = (512 + tiles - 1) / tiles;
   long long maxs = M >> 6;  // at least one 64-chunk per split
   if (want > maxs) want = maxs;
   if (const char* v = getenv("TDSA_GEMM_TN_SPLITS")) want = atoll(v);
