@@ -48,6 +48,30 @@ DEV_INLINE void tn_remap(int gx, int gy, int& tile, int& split, int tiles_k) {
   tile = tn * tiles_k + tk;
 }
 
+// Supertile remap (mode 2): group tiles into 2(tn) x 4(tk) rectangles and
+// place each rectangle's 8 tiles on ONE XCD (ids congruent mod 8 within a
+// 64-id window) — the XCD's L2 then re-serves both dy slices (2) and x
+// slices (4) instead of streaming 8 distinct slice pairs. Needs
+// tiles_n % 2 == 0, tiles_k % 4 == 0 and (supertiles * splits) % 8 == 0.
+DEV_INLINE void tn_remap2(int gx, int gy, int& tile, int& split,
+                          int tiles_k) {
+  const int tiles_n = gx / tiles_k;
+  if (tiles_n % 2 || tiles_k % 4) return;
+  const int sn = tiles_n >> 1, sk = tiles_k >> 2;
+  if ((sn * sk * gy) % 8) return;
+  const long long id = (long long)split * gx + tile;
+  const long long w = id >> 6;        // 64-id window
+  const int s_local = (int)(id & 7);  // supertile within window (XCD)
+  const int t_local = (int)((id & 63) >> 3);  // tile within supertile
+  const long long sup = w * 8 + s_local;
+  const int per_split = sn * sk;
+  split = (int)(sup / per_split);
+  const int ss = (int)(sup % per_split);
+  const int tn = (ss / sk) * 2 + (t_local >> 2);
+  const int tk = (ss % sk) * 4 + (t_local & 3);
+  tile = tn * tiles_k + tk;
+}
+
 // NW waves; output tile (NW*32) x 128. Wave w owns n-band [32w, 32w+32) x
 // 128 k. NW=8 halves the staged bytes per FLOP vs NW=4 (the dy slice is
 // amortized over twice the MFMA work) — used when N % 256 == 0.
@@ -60,7 +84,10 @@ __global__ void gemm_tn_kernel(const bf16* __restrict__ dy,
   const int tiles_k = K >> 7;
   int tile = blockIdx.x;
   int split = blockIdx.y;
-  if (remap) tn_remap(gridDim.x, gridDim.y, tile, split, tiles_k);
+  if (remap == 2)
+    tn_remap2(gridDim.x, gridDim.y, tile, split, tiles_k);
+  else if (remap)
+    tn_remap(gridDim.x, gridDim.y, tile, split, tiles_k);
   const int tn = tile / tiles_k;
   const int tk = tile % tiles_k;
   const int n0 = tn * (NW * 32);
