@@ -220,7 +220,10 @@ hipError_t tdsa_gemm_tn(const void* dy, const void* x, float* dw, long long M,
   const int nw = tn_nw(N);
   const long long chunks_per_split = ((M >> 6) + splits - 1) / splits;
   const long long m_per_split = chunks_per_split << 6;
-  int remap = 0;  // measured: no effect on any shape (see profiles/)
+  // mode 2 (supertile) measured +1-5% across shapes; mode 1 (tn-group)
+  // neutral — the chip-wide Infinity Cache already serves shared slices,
+  // so XCD placement only trims the L2-miss tail
+  int remap = 2;
   if (const char* v = getenv("TDSA_GEMM_TN_REMAP")) remap = atoi(v);
   dim3 grid((N / (nw * 32)) * (K >> 7), splits);
 #define TN_LAUNCH(NWV, AT)                                                  \
