@@ -50,11 +50,13 @@ hipError_t tdsa_sgd_step(void*, const void*, float*, float*, int, int, float,
                          int, hipStream_t);
 hipError_t tdsa_attn_fwd(const void*, const void*, const void*, void*, float*,
                          long long, long long, int, float, const long long*,
-                         const long long*, hipStream_t);
+                         const long long*, float, unsigned long long,
+                         hipStream_t);
 hipError_t tdsa_attn_bwd(const void*, const void*, const void*, const void*,
                          const float*, const void*, void*, void*, void*, float*,
                          long long, long long, int, float, const long long*,
-                         const long long*, const long long*, hipStream_t);
+                         const long long*, const long long*, float,
+                         unsigned long long, hipStream_t);
 int tdsa_gemm_tn_splits(long long M, int N, int K);
 hipError_t tdsa_gemm_tn(const void*, const void*, float*, long long, int, int,
                         hipStream_t);
@@ -435,7 +437,8 @@ static std::array<long long, 3> strides3(const at::Tensor& t) {
 
 std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                       double scale,
-                                      c10::optional<at::Tensor> out) {
+                                      c10::optional<at::Tensor> out,
+                                      double dropout_p, int64_t seed) {
   check_attn_tensor(q, "q");
   check_attn_tensor(k, "k");
   check_attn_tensor(v, "v");
@@ -449,9 +452,11 @@ std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   auto lse = at::empty({B, H, (long long)T}, q.options().dtype(at::kFloat));
   auto sq = strides3(q);
   auto so = strides3(o);
+  TORCH_CHECK(dropout_p >= 0.0 && dropout_p < 1.0, "bad dropout_p");
   check_hip(tdsa_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                           o.data_ptr(), lse.data_ptr<float>(), B, H, T,
-                          (float)scale, sq.data(), so.data(), cur_stream()),
+                          (float)scale, sq.data(), so.data(), (float)dropout_p,
+                          (unsigned long long)seed, cur_stream()),
             "attention_fwd");
   return {o, lse};
 }
@@ -461,7 +466,8 @@ std::vector<at::Tensor> attention_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                       at::Tensor dout, double scale,
                                       c10::optional<at::Tensor> dq_out,
                                       c10::optional<at::Tensor> dk_out,
-                                      c10::optional<at::Tensor> dv_out) {
+                                      c10::optional<at::Tensor> dv_out,
+                                      double dropout_p, int64_t seed) {
   check_attn_tensor(q, "q");
   check_attn_tensor(k, "k");
   check_attn_tensor(v, "v");
@@ -486,7 +492,8 @@ std::vector<at::Tensor> attention_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
                           o.data_ptr(), lsec.data_ptr<float>(), dout.data_ptr(),
                           dq.data_ptr(), dk.data_ptr(), dv.data_ptr(),
                           delta.data_ptr<float>(), B, H, T, (float)scale,
-                          sq.data(), so.data(), sd.data(), cur_stream()),
+                          sq.data(), so.data(), sd.data(), (float)dropout_p,
+                          (unsigned long long)seed, cur_stream()),
             "attention_bwd");
   return {dq, dk, dv};
 }
@@ -556,9 +563,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("sgd_step", &sgd_step);
   mod.def("sgd_step_multi", &sgd_step_multi);
   mod.def("attention_fwd", &attention_fwd, py::arg("q"), py::arg("k"),
-          py::arg("v"), py::arg("scale"), py::arg("out") = py::none());
+          py::arg("v"), py::arg("scale"), py::arg("out") = py::none(),
+          py::arg("dropout_p") = 0.0, py::arg("seed") = 0);
   mod.def("attention_bwd", &attention_bwd, py::arg("q"), py::arg("k"),
           py::arg("v"), py::arg("o"), py::arg("lse"), py::arg("dout"),
           py::arg("scale"), py::arg("dq") = py::none(),
-          py::arg("dk") = py::none(), py::arg("dv") = py::none());
+          py::arg("dk") = py::none(), py::arg("dv") = py::none(),
+          py::arg("dropout_p") = 0.0, py::arg("seed") = 0);
 }

@@ -78,6 +78,18 @@ struct GStride {
   int t;
 };
 
+// Counter-based dropout RNG (splitmix64 finalizer): keyed by
+// (bh, qrow, key) so the backward kernels REGENERATE the forward's mask
+// from (seed, indices) — nothing is stored, matching the recompute design.
+// keep <=> rng < threshold, threshold = (1-p) * 2^32.
+DEV_INLINE bool drop_keep(unsigned long long seed, unsigned long long idx,
+                          unsigned thr) {
+  unsigned long long z = seed + idx * 0x9E3779B97F4A7C15ull;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  return (unsigned)(z >> 32) < thr;
+}
+
 // C-layout -> A-fragment repack. Values live per lane as 16 C registers per
 // 32-wide tile (packed to bf16 word pairs wA[r1]=(r0=0,1), wB[r1]=(r0=2,3)).
 // A-frag slice s (k = 16s + 8*h32 + e at this lane's own 32-axis index)
@@ -122,16 +134,20 @@ DEV_INLINE bfrag load_frag(const bf16* p, int row, int col, int st) {
 // ---------------------------------------------------------------------------
 // Forward
 // ---------------------------------------------------------------------------
-template <int NW, int MINW = (NW == 8 ? 4 : 2)>
+template <int NW, bool DROP = false, int MINW = (NW == 8 ? 4 : 2)>
 // NW=8: cap at 128 VGPR so two 8-wave blocks are resident per CU (at 132
 // VGPR the 8-wave granularity rounds occupancy down to ONE block).
+// DROP: fused attention dropout — the row sum l_run uses the UNMASKED
+// exponentials (normalization is the true softmax sum) while the PV
+// accumulation consumes masked/rescaled P.
 __launch_bounds__(NW * WAVE, MINW)
 __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
                                 const bf16* __restrict__ k,
                                 const bf16* __restrict__ v,
                                 bf16* __restrict__ o, float* __restrict__ lse,
                                 int T, int H, float scale, GStride sq,
-                                GStride so) {
+                                GStride so, unsigned long long seed,
+                                unsigned keep_thr, float inv_keep) {
   constexpr int BM = NW * 32;
   constexpr int NT = NW * WAVE;
   __shared__ __attribute__((aligned(16))) char smem[2 * KVB * D * 2];
@@ -235,13 +251,21 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
     for (int t2 = 0; t2 < 2; ++t2)
 #pragma unroll
       for (int r1 = 0; r1 < 4; ++r1) {
-        const float p0 = fast_exp2(st[t2][4 * r1 + 0] - m_new);
-        const float p1 = fast_exp2(st[t2][4 * r1 + 1] - m_new);
-        const float p2 = fast_exp2(st[t2][4 * r1 + 2] - m_new);
-        const float p3 = fast_exp2(st[t2][4 * r1 + 3] - m_new);
-        psum += p0 + p1 + p2 + p3;
-        P.wA[t2][r1] = pack2(p0, p1);
-        P.wB[t2][r1] = pack2(p2, p3);
+        float p[4];
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          float pe = fast_exp2(st[t2][4 * r1 + e] - m_new);
+          psum += pe;
+          if (DROP) {
+            const int key = key0 + t2 * 32 + crow(4 * r1 + e, h32);
+            const unsigned long long idx =
+                ((unsigned long long)bh * T + row_me) * T + key;
+            pe = drop_keep(seed, idx, keep_thr) ? pe * inv_keep : 0.f;
+          }
+          p[e] = pe;
+        }
+        P.wA[t2][r1] = pack2(p[0], p[1]);
+        P.wB[t2][r1] = pack2(p[2], p[3]);
       }
     psum += __shfl_xor(psum, 32, WAVE);
     l_run = l_run * alpha + psum;
@@ -298,7 +322,7 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
 // ---------------------------------------------------------------------------
 // Backward dK/dV: one workgroup per (NW*32)-key block; wave w owns 32 keys.
 // ---------------------------------------------------------------------------
-template <int NW>
+template <int NW, bool DROP = false>
 __launch_bounds__(NW * WAVE)
 __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
                                     const bf16* __restrict__ k,
@@ -308,7 +332,9 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
                                     const float* __restrict__ delta,
                                     bf16* __restrict__ dk, bf16* __restrict__ dv,
                                     int T, int H, float scale, GStride sq,
-                                    GStride so, GStride sd) {
+                                    GStride so, GStride sd,
+                                    unsigned long long seed, unsigned keep_thr,
+                                    float inv_keep) {
   constexpr int BK = NW * 32;
   constexpr int NT = NW * WAVE;
   __shared__ __attribute__((aligned(16))) char smem[2 * KVB * D * 2 + 2 * KVB * 4];
@@ -409,8 +435,17 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
             const int lrow = t2 * 32 + crow(r, h32);
             float pp = fast_exp2(s_acc[r] - lds_lse[lrow]);
             if (MASK.value) pp = (key_me <= q0 + lrow) ? pp : 0.f;
-            p[e] = pp;
-            dsv[e] = pp * (dp_acc[r] - lds_dlt[lrow]);  // scale in epilogue
+            float pv = pp;                 // masked/rescaled P feeds dV
+            float dpv = dp_acc[r];         // masked/rescaled dP feeds dS
+            if (DROP) {
+              const unsigned long long idx =
+                  ((unsigned long long)bh * T + (q0 + lrow)) * T + key_me;
+              const bool keep = drop_keep(seed, idx, keep_thr);
+              pv = keep ? pp * inv_keep : 0.f;
+              dpv = keep ? dpv * inv_keep : 0.f;
+            }
+            p[e] = pv;
+            dsv[e] = pp * (dpv - lds_dlt[lrow]);  // scale in epilogue
           }
           P.wA[t2][r1] = pack2(p[0], p[1]);
           P.wB[t2][r1] = pack2(p[2], p[3]);
@@ -464,7 +499,7 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
 // ---------------------------------------------------------------------------
 // Backward dQ: one workgroup per (NW*32)-row Q block; wave w owns 32 rows.
 // ---------------------------------------------------------------------------
-template <int NW>
+template <int NW, bool DROP = false>
 __launch_bounds__(NW * WAVE)  // capping at 128 VGPR spills 164 B/lane here
 __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
                                    const bf16* __restrict__ k,
@@ -475,7 +510,8 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
                                    float* __restrict__ delta,
                                    bf16* __restrict__ dq, int T, int H,
                                    float scale, GStride sq, GStride so,
-                                   GStride sd) {
+                                   GStride sd, unsigned long long seed,
+                                   unsigned keep_thr, float inv_keep) {
   constexpr int BM = NW * 32;
   constexpr int NT = NW * WAVE;
   __shared__ __attribute__((aligned(16))) char smem[2 * KVB * D * 2];
@@ -583,7 +619,14 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
             const int key = key0 + t2 * 32 + crow(r, h32);
             pp = (key <= row_me) ? pp : 0.f;
           }
-          dsv[r] = pp * (dp_acc[r] - dlt_me);
+          float dpv = dp_acc[r];
+          if (DROP) {
+            const int key = key0 + t2 * 32 + crow(r, h32);
+            const unsigned long long idx =
+                ((unsigned long long)bh * T + row_me) * T + key;
+            dpv = drop_keep(seed, idx, keep_thr) ? dpv * inv_keep : 0.f;
+          }
+          dsv[r] = pp * (dpv - dlt_me);
         }
       };
       if (diag)
@@ -643,18 +686,35 @@ static int env_nw(const char* name, int dflt) {
 
 // strides arrays: {b, h, t} in elements, per tensor group:
 // sq = q/k/v, so = o (and dO in bwd), sd = dq/dk/dv.
+// dropout_p > 0 enables the fused-dropout template (seed regenerates the
+// identical mask in the backward kernels).
 hipError_t tdsa_attn_fwd(const void* q, const void* k, const void* v, void* o,
                          float* lse, long long B, long long H, int T,
                          float scale, const long long* sq_in,
-                         const long long* so_in, hipStream_t stream) {
+                         const long long* so_in, float dropout_p,
+                         unsigned long long seed, hipStream_t stream) {
   if (T % KVB) return hipErrorInvalidValue;
   GStride sq{sq_in[0], sq_in[1], (int)sq_in[2]};
   GStride so{so_in[0], so_in[1], (int)so_in[2]};
+  const bool drop = dropout_p > 0.0f;
+  const double keepd = 1.0 - (double)dropout_p;
+  const unsigned keep_thr = (unsigned)(keepd * 4294967296.0);
+  const float inv_keep = (float)(1.0 / keepd);
 #define LAUNCH_FWD(NW, MINW)                                                  \
-  hipLaunchKernelGGL((attn_fwd_kernel<NW, MINW>), dim3(T / (NW * 32), B * H), \
-                     dim3(NW * WAVE), 0, stream, (const bf16*)q,              \
-                     (const bf16*)k, (const bf16*)v, (bf16*)o, lse, T,        \
-                     (int)H, scale, sq, so)
+  do {                                                                        \
+    if (drop)                                                                 \
+      hipLaunchKernelGGL((attn_fwd_kernel<NW, true, MINW>),                   \
+                         dim3(T / (NW * 32), B * H), dim3(NW * WAVE), 0,      \
+                         stream, (const bf16*)q, (const bf16*)k,              \
+                         (const bf16*)v, (bf16*)o, lse, T, (int)H, scale,     \
+                         sq, so, seed, keep_thr, inv_keep);                   \
+    else                                                                      \
+      hipLaunchKernelGGL((attn_fwd_kernel<NW, false, MINW>),                  \
+                         dim3(T / (NW * 32), B * H), dim3(NW * WAVE), 0,      \
+                         stream, (const bf16*)q, (const bf16*)k,              \
+                         (const bf16*)v, (bf16*)o, lse, T, (int)H, scale,     \
+                         sq, so, seed, keep_thr, inv_keep);                   \
+  } while (0)
   const int nw_fwd = env_nw("TDSA_ATTN_FWD_NW", 8);
   const char* mv = getenv("TDSA_ATTN_FWD_MINW");
   const bool relaxed = mv && atoi(mv) == 3;
@@ -673,25 +733,37 @@ hipError_t tdsa_attn_bwd(const void* q, const void* k, const void* v,
                          void* dq, void* dk, void* dv, float* delta,
                          long long B, long long H, int T, float scale,
                          const long long* sq_in, const long long* so_in,
-                         const long long* sd_in, hipStream_t stream) {
+                         const long long* sd_in, float dropout_p,
+                         unsigned long long seed, hipStream_t stream) {
   if (T % KVB) return hipErrorInvalidValue;
   GStride sq{sq_in[0], sq_in[1], (int)sq_in[2]};
   GStride so{so_in[0], so_in[1], (int)so_in[2]};
   GStride sd{sd_in[0], sd_in[1], (int)sd_in[2]};
   const long long BH = B * H;
+  const bool drop = dropout_p > 0.0f;
+  const double keepd = 1.0 - (double)dropout_p;
+  const unsigned keep_thr = (unsigned)(keepd * 4294967296.0);
+  const float inv_keep = (float)(1.0 / keepd);
   // dq runs FIRST: it computes and publishes delta = rowsum(dO*O) from
   // fragments it loads anyway; dkv (same stream) consumes it.
-#define LAUNCH_BWD(NW)                                                        \
+#define LAUNCH_BWD_D(NW, DR)                                                  \
   do {                                                                        \
     dim3 grid(T / (NW * 32), BH);                                             \
-    hipLaunchKernelGGL(attn_bwd_dq_kernel<NW>, grid, dim3(NW * WAVE), 0,      \
-                       stream, (const bf16*)q, (const bf16*)k,                \
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<NW, DR>), grid, dim3(NW * WAVE),   \
+                       0, stream, (const bf16*)q, (const bf16*)k,             \
                        (const bf16*)v, (const bf16*)o, (const bf16*)dout,     \
-                       lse, delta, (bf16*)dq, T, (int)H, scale, sq, so, sd);  \
-    hipLaunchKernelGGL(attn_bwd_dkv_kernel<NW>, grid, dim3(NW * WAVE), 0,     \
-                       stream, (const bf16*)q, (const bf16*)k,                \
+                       lse, delta, (bf16*)dq, T, (int)H, scale, sq, so, sd,   \
+                       seed, keep_thr, inv_keep);                             \
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<NW, DR>), grid, dim3(NW * WAVE),  \
+                       0, stream, (const bf16*)q, (const bf16*)k,             \
                        (const bf16*)v, (const bf16*)dout, lse, delta,         \
-                       (bf16*)dk, (bf16*)dv, T, (int)H, scale, sq, so, sd);   \
+                       (bf16*)dk, (bf16*)dv, T, (int)H, scale, sq, so, sd,    \
+                       seed, keep_thr, inv_keep);                             \
+  } while (0)
+#define LAUNCH_BWD(NW)                                                        \
+  do {                                                                        \
+    if (drop) LAUNCH_BWD_D(NW, true);                                         \
+    else LAUNCH_BWD_D(NW, false);                                             \
   } while (0)
   // 8-wave blocks measured 285us vs 332us for 4-wave at B8/H16/T1024
   // (pre-tr16); staging amortization across 8 waves beat block overlap.
