@@ -178,7 +178,7 @@ def _attn_ref(q, k, v, scale):
     return torch.softmax(s, dim=-1) @ vf
 
 
-@pytest.mark.parametrize("T", [128, 192, 256])  # covers NW=4, NW=2, NW=8
+@pytest.mark.parametrize("T", [128, 192, 256, 2048])  # NW=4/2/8 + long context (2x the reference block_size cap)
 def test_attention_fwd_gpu(T):
     torch.manual_seed(0)
     B, H, D = 2, 3, 64
@@ -197,7 +197,7 @@ def test_attention_fwd_gpu(T):
     _close(lse, lse_ref, 2e-2, "attn lse")
 
 
-@pytest.mark.parametrize("T", [128, 192, 256])
+@pytest.mark.parametrize("T", [128, 192, 256, 2048])
 def test_attention_bwd_gpu(T):
     torch.manual_seed(1)
     B, H, D = 2, 2, 64
