@@ -323,6 +323,9 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
 // Backward dK/dV: one workgroup per (NW*32)-key block; wave w owns 32 keys.
 // ---------------------------------------------------------------------------
 template <int NW, bool DROP = false>
+// unconstrained VGPRs: the compiler takes ~256 (2 waves/SIMD) — capping to
+// 128 (MINW 4) spills 800 B/lane and 170 (MINW 3) still 288 B/lane, and the
+// measured time at 256 equals the kernel's best (169.8 us at B8)
 __launch_bounds__(NW * WAVE)
 __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
                                     const bf16* __restrict__ k,
