@@ -93,3 +93,13 @@ def test_loss_parity_world3(strategy):
         assert losses == pytest.approx(expected, abs=2e-4), (
             f"{strategy} rank {rank}: {losses} != {expected}"
         )
+
+
+def test_loss_parity_world8_zero2():
+    """Full-node world size (the driver's 8-GPU SCALE shape), gloo: the
+    partition table, per-tensor reduces and bucketed broadcasts must hold
+    parity at dp8 exactly as the real run will issue them."""
+    expected = W.single_device_losses()
+    results = run_distributed(W.train_strategy, world=8, args=("zero2",))
+    for rank, losses in results.items():
+        assert losses == pytest.approx(expected, abs=2e-4), (rank, losses)
