@@ -701,7 +701,11 @@ hipError_t tdsa_attn_fwd(const void* q, const void* k, const void* v, void* o,
   GStride so{so_in[0], so_in[1], (int)so_in[2]};
   const bool drop = dropout_p > 0.0f;
   const double keepd = 1.0 - (double)dropout_p;
-  const unsigned keep_thr = (unsigned)(keepd * 4294967296.0);
+  // clamp: for p -> 0, keepd*2^32 == 2^32 overflows (unsigned) to 0 and
+  // would drop EVERYTHING instead of nothing
+  const double thr_d = keepd * 4294967296.0;
+  const unsigned keep_thr =
+      thr_d >= 4294967295.0 ? 0xFFFFFFFFu : (unsigned)thr_d;
   const float inv_keep = (float)(1.0 / keepd);
 #define LAUNCH_FWD(NW, MINW)                                                  \
   do {                                                                        \
@@ -745,7 +749,11 @@ hipError_t tdsa_attn_bwd(const void* q, const void* k, const void* v,
   const long long BH = B * H;
   const bool drop = dropout_p > 0.0f;
   const double keepd = 1.0 - (double)dropout_p;
-  const unsigned keep_thr = (unsigned)(keepd * 4294967296.0);
+  // clamp: for p -> 0, keepd*2^32 == 2^32 overflows (unsigned) to 0 and
+  // would drop EVERYTHING instead of nothing
+  const double thr_d = keepd * 4294967296.0;
+  const unsigned keep_thr =
+      thr_d >= 4294967295.0 ? 0xFFFFFFFFu : (unsigned)thr_d;
   const float inv_keep = (float)(1.0 / keepd);
   // dq runs FIRST: it computes and publishes delta = rowsum(dO*O) from
   // fragments it loads anyway; dkv (same stream) consumes it.
