@@ -73,3 +73,26 @@ def test_zero2flat_incomplete_backward_raises():
     opt = tdsa.Zero2FlatAdamW(model, lr=1e-3)
     with pytest.raises(RuntimeError, match="never completed"):
         opt.step()  # no backward ran
+
+
+def test_zero2flat_sgd_world1_parity():
+    def run(flat):
+        torch.manual_seed(0)
+        m = GPT2Model(w.make_cfg())
+        if flat:
+            model = tdsa.Zero2Flat(m)
+            opt = tdsa.Zero2FlatSGD(model, lr=1e-2, momentum=0.9)
+        else:
+            model = tdsa.Single(m)
+            opt = tdsa.SGD(model.named_parameters(), lr=1e-2, momentum=0.9)
+        x, y = w.batch()
+        losses = []
+        for _ in range(3):
+            model.require_backward_grad_sync = True
+            _, loss = model(x, y)
+            loss.backward()
+            opt.step()
+            losses.append(loss.item())
+        return losses
+
+    assert run(True) == pytest.approx(run(False), rel=1e-5)
