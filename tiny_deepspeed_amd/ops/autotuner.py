@@ -138,7 +138,12 @@ def default_tuner():
             else:
                 import atexit
 
-                atexit.register(
-                    lambda: _DEFAULT_TUNER and _DEFAULT_TUNER.save_cache(cache)
-                )
+                def _save():
+                    try:
+                        if _DEFAULT_TUNER is not None:
+                            _DEFAULT_TUNER.save_cache(cache)
+                    except OSError:
+                        pass  # cache is an optimization; never fail exit
+
+                atexit.register(_save)
     return _DEFAULT_TUNER
