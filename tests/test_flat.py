@@ -96,3 +96,48 @@ def test_zero2flat_sgd_world1_parity():
         return losses
 
     assert run(True) == pytest.approx(run(False), rel=1e-5)
+
+
+def test_zero2flat_checkpoint_same_world(tmp_path):
+    """Flat-mode checkpoints restore at the same world size (numel-shard
+    state; cross-world resharding is the per-tensor mode's job and the
+    loader REPORTS rather than guesses on mismatch)."""
+    import os
+
+    from tiny_deepspeed_amd.utils.checkpoint import (save_checkpoint,
+                                                     load_checkpoint)
+
+    def build():
+        torch.manual_seed(0)
+        model = tdsa.Zero2Flat(GPT2Model(w.make_cfg()))
+        opt = tdsa.Zero2FlatAdamW(model, lr=1e-3)
+        return model, opt
+
+    model, opt = build()
+    x, y = w.batch()
+    for _ in range(2):
+        model.require_backward_grad_sync = True
+        _, loss = model(x, y)
+        loss.backward()
+        opt.step()
+    path = os.path.join(tmp_path, "flat.pt")
+    save_checkpoint(path, model, opt, step=2)
+    ref = []
+    for _ in range(2):
+        model.require_backward_grad_sync = True
+        _, loss = model(x, y)
+        loss.backward()
+        opt.step()
+        ref.append(loss.item())
+
+    model2, opt2 = build()
+    step, report = load_checkpoint(path, model2, opt2, return_report=True)
+    assert step == 2 and report.ok()
+    out = []
+    for _ in range(2):
+        model2.require_backward_grad_sync = True
+        _, loss = model2(x, y)
+        loss.backward()
+        opt2.step()
+        out.append(loss.item())
+    assert out == pytest.approx(ref, rel=1e-5)
