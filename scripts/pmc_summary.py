@@ -16,13 +16,25 @@ def main():
         "SELECT name FROM sqlite_master WHERE type='table' AND name LIKE "
         "'rocpd_pmc_event%'")][0]
     sfx = t.replace("rocpd_pmc_event_", "")
-    # discover join columns defensively
+    # discover join columns defensively (schema varies across rocprofv3)
     cols = [r[1] for r in cur.execute(f"PRAGMA table_info({t})")]
+    kd_cols = [r[1] for r in cur.execute(
+        f"PRAGMA table_info(rocpd_kernel_dispatch_{sfx})")]
+    pmc_id = next((c for c in cols if "pmc" in c.lower()), None)
+    disp = next((c for c in cols if "dispatch" in c.lower()
+                 or "event" in c.lower() or "corr" in c.lower()), None)
+    kd_key = "id"
+    if disp and "corr" in disp.lower():
+        kd_key = next((c for c in kd_cols if "corr" in c.lower()), "id")
+    if pmc_id is None or disp is None:
+        print("pmc_event columns:", cols)
+        print("kernel_dispatch columns:", kd_cols)
+        raise SystemExit("cannot infer join columns")
     rows = list(cur.execute(f"""
         SELECT ks.display_name, pi.name, SUM(pe.value)
         FROM {t} pe
-        JOIN rocpd_info_pmc_{sfx} pi ON pe.pmc_id = pi.id
-        JOIN rocpd_kernel_dispatch_{sfx} kd ON pe.dispatch_id = kd.id
+        JOIN rocpd_info_pmc_{sfx} pi ON pe.{pmc_id} = pi.id
+        JOIN rocpd_kernel_dispatch_{sfx} kd ON pe.{disp} = kd.{kd_key}
         JOIN rocpd_info_kernel_symbol_{sfx} ks ON kd.kernel_id = ks.id
         GROUP BY ks.display_name, pi.name"""))
     per_kernel = collections.defaultdict(dict)
