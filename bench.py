@@ -170,6 +170,17 @@ def main():
     n_gpus = world_size if have_gpu else args.gpus
     tokens_per_step = args.batch * args.seq * world_size
     value = tokens_per_step * args.steps / elapsed
+    if distributed and rank == 0:
+        # diagnostic for multi-GPU runs (stderr; the stdout JSON contract
+        # is untouched): per-step collective counts and payload GB
+        comm = getattr(wrapped, "comm", None)
+        if comm is not None and comm.stats:
+            total = args.warmup + args.steps
+            stats = {k: {"per_step": round(v[0] / total, 1),
+                         "gb_per_step": round(v[1] / total / 2**30, 3)}
+                     for k, v in sorted(comm.stats.items())}
+            print(f"[comm-stats world={world_size}] {json.dumps(stats)}",
+                  file=sys.stderr)
     if rank == 0:
         out = {
             "metric": "tokens/sec (whole node) GPT-2 ZeRO-2",

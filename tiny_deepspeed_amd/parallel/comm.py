@@ -73,6 +73,14 @@ class CommContext:
         # gloo flat broadcasts whose scatter-back is deferred to sync():
         # (work, flat, bucket, numels) per channel
         self._pending_flat = {REDUCE: [], GATHER: []}
+        # observability: per-collective launch counts and payload bytes
+        # (read by bench.py's comm-stats line; negligible overhead)
+        self.stats = {}
+
+    def _count(self, kind, t):
+        c = self.stats.setdefault(kind, [0, 0])
+        c[0] += 1
+        c[1] += t.numel() * t.element_size()
 
     # ------------------------------------------------------------------ #
     def _inactive(self):
@@ -140,6 +148,7 @@ class CommContext:
         """Async in-place average-all-reduce; returns t."""
         if self._inactive():
             return t
+        self._count("all_reduce", t)
 
         def run(async_op=False):
             t.div_(self.world_size)
@@ -153,6 +162,7 @@ class CommContext:
         """Async in-place average-reduce to `owner`; valid only there."""
         if self._inactive():
             return t
+        self._count("reduce", t)
 
         def run(async_op=False):
             t.div_(self.world_size)
@@ -166,6 +176,7 @@ class CommContext:
         """Async in-place broadcast from `src`; returns t."""
         if self._inactive():
             return t
+        self._count("broadcast", t)
 
         def run(async_op=False):
             return dist.broadcast(t, src=src, group=self.pg[channel],
@@ -244,6 +255,7 @@ class CommContext:
         if self._inactive():
             out_shard.copy_(in_flat[: out_shard.numel()])
             return out_shard
+        self._count("reduce_scatter", in_flat)
 
         def run(async_op=False):
             in_flat.div_(self.world_size)
@@ -273,6 +285,7 @@ class CommContext:
             if out_flat[:n].data_ptr() != in_shard.data_ptr():
                 out_flat[:n].copy_(in_shard)
             return out_flat
+        self._count("all_gather", out_flat)
 
         def run(async_op=False):
             if dist.get_backend(self.pg[REDUCE]) == "gloo":
