@@ -78,16 +78,24 @@ struct GStride {
   int t;
 };
 
-// Counter-based dropout RNG (splitmix64 finalizer): keyed by
-// (bh, qrow, key) so the backward kernels REGENERATE the forward's mask
-// from (seed, indices) — nothing is stored, matching the recompute design.
+// Counter-based dropout RNG: keyed by (bh, qrow, key) so the backward
+// kernels REGENERATE the forward's mask from (seed, indices) — nothing is
+// stored, matching the recompute design. 32-bit fold + murmur3 fmix32
+// finalizer: the first splitmix64 version kept 64-bit temporaries live
+// across the softmax loop and pushed the DROP template into a 312 B/lane
+// spill — the avalanche quality of fmix32 is ample for dropout.
 // keep <=> rng < threshold, threshold = (1-p) * 2^32.
 DEV_INLINE bool drop_keep(unsigned long long seed, unsigned long long idx,
                           unsigned thr) {
-  unsigned long long z = seed + idx * 0x9E3779B97F4A7C15ull;
-  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
-  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
-  return (unsigned)(z >> 32) < thr;
+  unsigned h = (unsigned)(idx ^ (idx >> 32));
+  h += (unsigned)seed;
+  h ^= (unsigned)(seed >> 32);
+  h ^= h >> 16;
+  h *= 0x85EBCA6Bu;
+  h ^= h >> 13;
+  h *= 0xC2B2AE35u;
+  h ^= h >> 16;
+  return h < thr;
 }
 
 // C-layout -> A-fragment repack. Values live per lane as 16 C registers per
@@ -190,6 +198,9 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
 
   const int row_lo = qb * BM + w * 32;
   const int row_me = row_lo + q32;
+  // dropout index base (loop-invariant): idx = base + key
+  const unsigned long long drop_base =
+      DROP ? ((unsigned long long)bh * T + row_me) * T : 0;
   const int n_kv = (qb + 1) * BM / KVB;
   short8v pend_k[Stage<NT>::REPS], pend_v[Stage<NT>::REPS];
   stage_k.fetch(pend_k);
@@ -258,9 +269,9 @@ __global__ void attn_fwd_kernel(const bf16* __restrict__ q,
           psum += pe;
           if (DROP) {
             const int key = key0 + t2 * 32 + crow(4 * r1 + e, h32);
-            const unsigned long long idx =
-                ((unsigned long long)bh * T + row_me) * T + key;
-            pe = drop_keep(seed, idx, keep_thr) ? pe * inv_keep : 0.f;
+            pe = drop_keep(seed, drop_base + key, keep_thr)
+                     ? pe * inv_keep
+                     : 0.f;
           }
           p[e] = pe;
         }
@@ -375,6 +386,9 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
   const float kscale = scale * LOG2E;
   const int key_lo = jb * BK + w * 32;
   const int key_me = key_lo + k32;
+  // dropout index base: idx = base + qrow*T + key_me (qrow*T fits 32 bits)
+  const unsigned long long drop_bhTT =
+      DROP ? (unsigned long long)bh * T * T : 0;
   bfrag k_frag[4], v_frag[4];  // B operands: col = k32, k-slices over D
 #pragma unroll
   for (int s = 0; s < 4; ++s) {
@@ -441,9 +455,10 @@ __global__ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
             float pv = pp;                 // masked/rescaled P feeds dV
             float dpv = dp_acc[r];         // masked/rescaled dP feeds dS
             if (DROP) {
-              const unsigned long long idx =
-                  ((unsigned long long)bh * T + (q0 + lrow)) * T + key_me;
-              const bool keep = drop_keep(seed, idx, keep_thr);
+              const bool keep = drop_keep(
+                  seed,
+                  drop_bhTT + (unsigned)((q0 + lrow) * T) + key_me,
+                  keep_thr);
               pv = keep ? pp * inv_keep : 0.f;
               dpv = keep ? dpv * inv_keep : 0.f;
             }
@@ -557,6 +572,8 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   }
   const int row_lo = qb * BM + w * 32;
   const int row_me = row_lo + q32;
+  const unsigned long long drop_base =
+      DROP ? ((unsigned long long)bh * T + row_me) * T : 0;
   const float lse2_me = lse[bh * T + row_me] * LOG2E;
   // delta = rowsum(dO*O) computed here from the fragments already in
   // registers (+ one O load) and PUBLISHED for the dkv kernel, which is
@@ -625,9 +642,9 @@ __global__ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
           float dpv = dp_acc[r];
           if (DROP) {
             const int key = key0 + t2 * 32 + crow(r, h32);
-            const unsigned long long idx =
-                ((unsigned long long)bh * T + row_me) * T + key;
-            dpv = drop_keep(seed, idx, keep_thr) ? dpv * inv_keep : 0.f;
+            dpv = drop_keep(seed, drop_base + key, keep_thr)
+                      ? dpv * inv_keep
+                      : 0.f;
           }
           dsv[r] = pp * (dpv - dlt_me);
         }
@@ -724,7 +741,10 @@ hipError_t tdsa_attn_fwd(const void* q, const void* k, const void* v, void* o,
   } while (0)
   const int nw_fwd = env_nw("TDSA_ATTN_FWD_NW", 8);
   const char* mv = getenv("TDSA_ATTN_FWD_MINW");
-  const bool relaxed = mv && atoi(mv) == 3;
+  const bool relaxed = (mv && atoi(mv) == 3) || drop;
+  // dropout: the no-drop kernel already fills the 128-VGPR cap, so the
+  // DROP variant spills 316 B/lane under MINW=4 — the relaxed cap
+  // (168 VGPR, 1 block/CU) measured faster for it
   if (T % 256 == 0 && nw_fwd == 8) {
     if (relaxed) LAUNCH_FWD(8, 3);  // A/B: no 128-cap spill, 1 block/CU
     else LAUNCH_FWD(8, 4);
