@@ -87,3 +87,19 @@ def test_forced_ddp_training_matches_unforced(world1_pg):
     plain = _train_losses(False)
     assert forced == pytest.approx(plain, rel=0, abs=0), (forced, plain)
     assert forced[-1] < forced[0]
+
+
+def test_single_communicator_mode(world1_pg, monkeypatch):
+    """TDSA_COMM_SINGLE=1: both channels share one communicator and one
+    stream — the safety fallback for concurrent-communicator hazards."""
+    monkeypatch.setenv("TDSA_COMM_SINGLE", "1")
+    comm = CommContext()
+    assert comm.single
+    assert comm.pg["gather"] is comm.pg["reduce"]
+    t = torch.randn(32)
+    ref = t.clone()
+    comm.gather_broadcast(t, 0)
+    comm.wait_gather()
+    comm.all_reduce_avg(t)
+    comm.sync()
+    torch.testing.assert_close(t, ref)

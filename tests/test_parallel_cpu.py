@@ -103,3 +103,13 @@ def test_loss_parity_world8_zero2():
     results = run_distributed(W.train_strategy, world=8, args=("zero2",))
     for rank, losses in results.items():
         assert losses == pytest.approx(expected, abs=2e-4), (rank, losses)
+
+
+def test_zero3_parity_single_comm_mode(monkeypatch):
+    """ZeRO-3 under TDSA_COMM_SINGLE=1 (the concurrent-communicator
+    fallback): same loss parity as the dual-communicator default."""
+    monkeypatch.setenv("TDSA_COMM_SINGLE", "1")
+    expected = W.single_device_losses()
+    results = run_distributed(W.train_strategy, world=2, args=("zero3",))
+    for rank, losses in results.items():
+        assert losses == pytest.approx(expected, abs=2e-4), (rank, losses)

@@ -50,11 +50,19 @@ class CommContext:
         # early-returns below would otherwise skip (hardware evidence for
         # the multi-GPU machinery before an 8-GPU node exists).
         self.force = os.environ.get("TDSA_COMM_FORCE", "0") == "1"
+        # TDSA_COMM_SINGLE=1: one communicator + one comm stream for BOTH
+        # channels. Safety fallback for ZeRO-3 at scale: concurrent
+        # communicators can deadlock if their kernels interleave differently
+        # across ranks; with a single stream the enqueue order is the
+        # (identical) program order on every rank. Costs gather/reduce
+        # serialization — use only if the dual-communicator mode misbehaves.
+        self.single = os.environ.get("TDSA_COMM_SINGLE", "0") == "1"
         self.pg = {REDUCE: process_group, GATHER: process_group}
         if self.initialized:
             self.rank = dist.get_rank(process_group)
             self.world_size = dist.get_world_size(process_group)
-            if (self.world_size > 1 or self.force) and process_group is None:
+            if ((self.world_size > 1 or self.force) and process_group is None
+                    and not self.single):
                 # second communicator for the gather channel (must be
                 # constructed collectively, identical on all ranks)
                 self.pg[GATHER] = dist.new_group(backend=dist.get_backend())
@@ -64,9 +72,11 @@ class CommContext:
             self.force = False
         self.is_cuda = torch.cuda.is_available()
         use_streams = self.is_cuda and use_comm_stream
+        reduce_stream = torch.cuda.Stream() if use_streams else None
         self.streams = {
-            REDUCE: torch.cuda.Stream() if use_streams else None,
-            GATHER: torch.cuda.Stream() if use_streams else None,
+            REDUCE: reduce_stream,
+            GATHER: (reduce_stream if self.single else torch.cuda.Stream())
+                    if use_streams else None,
         }
         self._works = {REDUCE: [], GATHER: []}
         self._keepalive = []
