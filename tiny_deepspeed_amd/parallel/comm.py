@@ -71,7 +71,14 @@ class CommContext:
             self.world_size = 1
             self.force = False
         self.is_cuda = torch.cuda.is_available()
-        use_streams = self.is_cuda and use_comm_stream
+        # streams are keyed on the BACKEND, not on GPU visibility: a gloo
+        # process group on a GPU box moves CPU tensors, and record_stream
+        # on those raises (found running the combined CPU+GPU suite on an
+        # MI355X box). Uninitialized contexts keep streams harmless — every
+        # collective early-returns at _inactive().
+        backend = (str(dist.get_backend()) if self.initialized else None)
+        use_streams = (self.is_cuda and use_comm_stream
+                       and (backend is None or "nccl" in backend))
         reduce_stream = torch.cuda.Stream() if use_streams else None
         self.streams = {
             REDUCE: reduce_stream,
